@@ -1,0 +1,25 @@
+"""raft_amd — MI355X-native RAFT optical-flow framework.
+
+A from-scratch rebuild of the capabilities of gonglixue/RAFT-tf
+(TF-1.x/tensorpack inference-only RAFT) as an MI355X-first framework:
+
+  * PyTorch-ROCm at the framework layer (NCHW modules, eager + HIP-graph
+    capture of the recurrent update loop),
+  * hand-written CDNA4 (gfx950) HIP kernels for the four hot ops —
+    all-pairs correlation volume build (MFMA), correlation-pyramid pooling,
+    multi-scale 81-tap pyramid lookup, fused ConvGRU gate math — and the
+    8x convex upsample,
+  * RCCL-over-xGMI data parallelism (one process per GPU).
+
+Checkpoint contract: the reference's ``.npz`` layout (TF variable names,
+HWIO conv weights) is read and written by :mod:`raft_amd.utils.checkpoint`
+(reference: infer_raft.py:77, tensorpack ``get_model_loader``).
+
+The pure-PyTorch implementations in :mod:`raft_amd.ops.torch_ref` are
+numerics-exact to the reference graph (see each docstring for the
+file:line it mirrors) and serve as the golden oracle for HIP kernel tests.
+"""
+
+__version__ = "0.1.0"
+
+from raft_amd.models.raft import RAFT, RaftConfig  # noqa: F401

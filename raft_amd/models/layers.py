@@ -1,0 +1,87 @@
+"""Primitive layers: TF-"SAME"-padded conv and the norm dispatch.
+
+The reference builds every conv with tensorpack ``Conv2D(..., padding='same')``
+(networks/model_utils.py throughout).  TF SAME padding is *asymmetric* for
+stride > 1 (extra pixel on the bottom/right), unlike PyTorch's symmetric
+padding — for checkpoint fidelity with the reference's converted ``.npz``
+weights we replicate TF semantics exactly.
+
+Norm dispatch mirrors networks/model_utils.py:6-17:
+  'group'    -> GroupNorm(groups=C//8)    (reference's version is NCHW-broken
+                on its NHWC tensors; ours is correct NCHW — the configured
+                models never use 'group' so this is a strict improvement)
+  'batch'    -> BatchNorm (eps 1e-5, EMA momentum 0.9 ≡ torch momentum 0.1)
+  'instance' -> InstanceNorm with center=False scale=False (NO affine params,
+                model_utils.py:13) — this is why fnet checkpoints carry no
+                norm variables.
+  'none'     -> identity
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class Conv2dTF(nn.Conv2d):
+    """Conv2d with TensorFlow 'SAME' padding semantics.
+
+    For stride 1 and odd kernels this is plain symmetric padding (fast path,
+    padding folded into conv2d). For stride > 1, TF pads
+    ``total = max((ceil(in/s)-1)*s + k - in, 0)`` split floor/ceil
+    (beg = total//2, end = total - beg) — asymmetric when total is odd.
+    """
+
+    def __init__(self, *args, **kwargs):
+        kwargs.pop("padding", None)
+        super().__init__(*args, padding=0, **kwargs)
+
+    @staticmethod
+    def _same_pad(size: int, k: int, s: int) -> tuple[int, int]:
+        total = max((math.ceil(size / s) - 1) * s + k - size, 0)
+        return total // 2, total - total // 2
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        ih, iw = x.shape[-2:]
+        ph0, ph1 = self._same_pad(ih, self.kernel_size[0], self.stride[0])
+        pw0, pw1 = self._same_pad(iw, self.kernel_size[1], self.stride[1])
+        if ph0 == ph1 and pw0 == pw1:
+            return F.conv2d(x, self.weight, self.bias, self.stride,
+                            (ph0, pw0), self.dilation, self.groups)
+        x = F.pad(x, (pw0, pw1, ph0, ph1))
+        return F.conv2d(x, self.weight, self.bias, self.stride,
+                        0, self.dilation, self.groups)
+
+
+def make_norm(norm_fn: str, channels: int) -> nn.Module:
+    """Norm dispatch per networks/model_utils.py:6-17 (see module docstring)."""
+    if norm_fn == "group":
+        return nn.GroupNorm(channels // 8, channels, eps=1e-5)
+    if norm_fn == "batch":
+        return nn.BatchNorm2d(channels, eps=1e-5, momentum=0.1)
+    if norm_fn == "instance":
+        # tensorpack InstanceNorm(center=False, scale=False): no learnable
+        # affine, eps 1e-5 (model_utils.py:13).
+        return nn.InstanceNorm2d(channels, eps=1e-5, affine=False,
+                                 track_running_stats=False)
+    if norm_fn == "none":
+        return nn.Identity()
+    raise ValueError(f"unknown norm_fn {norm_fn!r}")
+
+
+def coords_grid(batch: int, ht: int, wd: int, device=None,
+                dtype=torch.float32) -> torch.Tensor:
+    """Pixel-coordinate grid ``[B, 2, H, W]`` with channel 0 = x, 1 = y.
+
+    Mirrors networks/utils.py:4-11 (x-major (x, y) last-dim there; we keep
+    NCHW with the same (x, y) channel order).
+    """
+    y, x = torch.meshgrid(
+        torch.arange(ht, device=device, dtype=dtype),
+        torch.arange(wd, device=device, dtype=dtype),
+        indexing="ij",
+    )
+    grid = torch.stack([x, y], dim=0)  # [2, H, W]
+    return grid.unsqueeze(0).expand(batch, -1, -1, -1).contiguous()

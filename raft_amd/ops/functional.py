@@ -1,0 +1,188 @@
+"""Autograd-wrapped hot ops: HIP kernels on GPU, torch_ref elsewhere.
+
+Each op dispatches on the input's device:
+  * CUDA/HIP tensor  -> hand-written gfx950 kernel from the in-tree
+    extension (required; loud failure if unbuilt — see raft_amd.ops).
+  * CPU tensor       -> raft_amd.ops.torch_ref (plain PyTorch, autograd-native).
+
+The HIP forward/backward pairs are wrapped in torch.autograd.Function so the
+training path (config 3, DP=8) backprops through them.  Plain GEMM backward
+passes go through torch.matmul (rocBLAS/hipBLASLt) — library GEMMs are the
+sanctioned path for non-fused matmuls.
+"""
+from __future__ import annotations
+
+import math
+import os
+from typing import List
+
+import torch
+
+from raft_amd.ops import torch_ref
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if os.environ.get("RAFT_AMD_FORCE_TORCH", "0") == "1":
+        return False
+    from raft_amd.ops import require_hip
+    require_hip()  # loud failure if the extension is missing on GPU
+    return True
+
+
+class _CorrVolume(torch.autograd.Function):
+    """C = F1·F2ᵀ/sqrt(c): HIP MFMA forward, torch.matmul backward."""
+
+    @staticmethod
+    def forward(ctx, fmap1, fmap2):
+        from raft_amd.ops import require_hip
+        corr = require_hip().corr_volume(fmap1.contiguous(), fmap2.contiguous())
+        ctx.save_for_backward(fmap1, fmap2)
+        return corr
+
+    @staticmethod
+    def backward(ctx, grad_corr):
+        fmap1, fmap2 = ctx.saved_tensors
+        B, C, H, W = fmap1.shape
+        scale = 1.0 / math.sqrt(C)
+        g = grad_corr.reshape(B, H * W, H * W).float() * scale
+        f1 = fmap1.reshape(B, C, H * W).float()
+        f2 = fmap2.reshape(B, C, H * W).float()
+        # dF1 = dC · F2 ; dF2 = dCᵀ · F1   (both [B, HW, C] -> back to NCHW)
+        g1 = torch.matmul(g, f2.transpose(1, 2))       # [B, HW, C]
+        g2 = torch.matmul(g.transpose(1, 2), f1.transpose(1, 2))
+        g1 = g1.transpose(1, 2).reshape(B, C, H, W).to(fmap1.dtype)
+        g2 = g2.transpose(1, 2).reshape(B, C, H, W).to(fmap2.dtype)
+        return g1, g2
+
+
+def corr_volume(fmap1: torch.Tensor, fmap2: torch.Tensor) -> torch.Tensor:
+    if _use_hip(fmap1):
+        return _CorrVolume.apply(fmap1, fmap2)
+    return torch_ref.corr_volume(fmap1, fmap2)
+
+
+class _CorrPool2x(torch.autograd.Function):
+    """2x2/2 avg-pool (TF VALID) over the target dims of the volume."""
+
+    @staticmethod
+    def forward(ctx, corr):
+        from raft_amd.ops import require_hip
+        ctx.in_shape = corr.shape
+        return require_hip().corr_pool2x(corr.contiguous())
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        B, HW, H2, W2 = ctx.in_shape
+        g = grad_out * 0.25
+        g = g.repeat_interleave(2, dim=-2).repeat_interleave(2, dim=-1)
+        if g.shape[-2] < H2 or g.shape[-1] < W2:  # VALID drops odd tails
+            pad_h = H2 - g.shape[-2]
+            pad_w = W2 - g.shape[-1]
+            g = torch.nn.functional.pad(g, (0, pad_w, 0, pad_h))
+        return g
+
+
+def corr_pyramid(fmap1: torch.Tensor, fmap2: torch.Tensor,
+                 num_levels: int = 4) -> List[torch.Tensor]:
+    """Build volume + pooled pyramid (model_utils.py:199-221)."""
+    corr = corr_volume(fmap1, fmap2)
+    if _use_hip(corr):
+        levels = [corr]
+        for _ in range(num_levels - 1):
+            levels.append(_CorrPool2x.apply(levels[-1]))
+        return levels
+    return torch_ref.corr_pyramid_pool(corr, num_levels)
+
+
+class _CorrLookup(torch.autograd.Function):
+    """(2r+1)^2-tap x L-level bilinear window gather (K3)."""
+
+    @staticmethod
+    def forward(ctx, coords, radius, *levels):
+        from raft_amd.ops import require_hip
+        out = require_hip().corr_lookup(list(levels), coords.contiguous(),
+                                        radius)
+        ctx.save_for_backward(coords)
+        ctx.radius = radius
+        ctx.level_shapes = [tuple(l.shape) for l in levels]
+        ctx.level_dtypes = [l.dtype for l in levels]
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        from raft_amd.ops import require_hip
+        (coords,) = ctx.saved_tensors
+        grads = require_hip().corr_lookup_backward(
+            grad_out.contiguous(), coords.contiguous(), ctx.radius,
+            [list(s) for s in ctx.level_shapes])
+        grads = [g.to(dt) for g, dt in zip(grads, ctx.level_dtypes)]
+        # coords are detached in the RAFT loop (RAFT.py:93) — no coord grad.
+        return (None, None, *grads)
+
+
+def corr_lookup(pyramid: List[torch.Tensor], coords: torch.Tensor,
+                radius: int) -> torch.Tensor:
+    if _use_hip(coords):
+        return _CorrLookup.apply(coords, radius, *pyramid)
+    return torch_ref.corr_lookup(pyramid, coords, radius)
+
+
+class _GruGates(torch.autograd.Function):
+    """Fused pointwise h' = (1-σ(z))h + σ(z)tanh(q) with analytic backward."""
+
+    @staticmethod
+    def forward(ctx, h, z_act, q_act):
+        from raft_amd.ops import require_hip
+        out = require_hip().gru_gates_fwd(h.contiguous(), z_act.contiguous(),
+                                          q_act.contiguous())
+        ctx.save_for_backward(h, z_act, q_act)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_h_new):
+        from raft_amd.ops import require_hip
+        h, z_act, q_act = ctx.saved_tensors
+        gh, gz, gq = require_hip().gru_gates_bwd(grad_h_new.contiguous(),
+                                                 h.contiguous(),
+                                                 z_act.contiguous(),
+                                                 q_act.contiguous())
+        return gh, gz, gq
+
+
+def gru_gates(h: torch.Tensor, z_act: torch.Tensor,
+              q_act: torch.Tensor) -> torch.Tensor:
+    if _use_hip(h):
+        return _GruGates.apply(h, z_act, q_act)
+    return torch_ref.gru_gates(h, z_act, q_act)
+
+
+class _ConvexUpsample(torch.autograd.Function):
+    """8x convex upsample (K5): softmax over 9 taps fused into the kernel."""
+
+    @staticmethod
+    def forward(ctx, flow, mask):
+        from raft_amd.ops import require_hip
+        out = require_hip().convex_upsample(flow.contiguous(),
+                                            mask.contiguous())
+        ctx.save_for_backward(flow, mask)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_up):
+        from raft_amd.ops import require_hip
+        flow, mask = ctx.saved_tensors
+        gf, gm = require_hip().convex_upsample_backward(
+            grad_up.contiguous(), flow.contiguous(), mask.contiguous())
+        return gf, gm
+
+
+def convex_upsample(flow: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+    if _use_hip(flow):
+        return _ConvexUpsample.apply(flow, mask)
+    return torch_ref.convex_upsample(flow, mask)
+
+
+def upflow8(flow: torch.Tensor) -> torch.Tensor:
+    return torch_ref.upflow8(flow)

@@ -1,0 +1,144 @@
+"""``.npz`` checkpoint compatibility — the reference's weight contract.
+
+The reference loads weights via tensorpack ``get_model_loader(path)``
+(infer_raft.py:77): a ``.npz`` archive of ``{TF_variable_name: ndarray}``
+with tensorpack naming — ``Conv2D`` -> ``W`` (HWIO layout) / ``b``,
+``BatchNorm`` -> ``gamma``/``beta``/``mean/EMA``/``variance/EMA`` —
+see SURVEY.md §5.4 for the full key tree.
+
+Because this package's module child names mirror the TF scopes
+(fnet/cnet/update_block/... — see models/encoders.py docstring), the
+mapping is mechanical:
+
+    TF  fnet/layer2/0/downsample.0/W   [kh,kw,cin,cout]  (HWIO)
+    PT  fnet.layer2.0.downsample.0.weight [cout,cin,kh,kw] (OIHW)
+
+plus leaf renames (W->weight, b->bias, gamma->weight, beta->bias,
+mean/EMA->running_mean, variance/EMA->running_var).
+"""
+from __future__ import annotations
+
+import io
+from typing import Dict
+
+import numpy as np
+import torch
+
+_NORM_LEAVES = {
+    "gamma": "weight",
+    "beta": "bias",
+    "mean/EMA": "running_mean",
+    "variance/EMA": "running_var",
+}
+
+
+def tf_key_to_torch(key: str):
+    """Map a TF variable name to (torch state_dict key, needs_transpose)."""
+    key = key[:-2] if key.endswith(":0") else key
+    for tf_leaf, pt_leaf in _NORM_LEAVES.items():
+        if key.endswith("/" + tf_leaf):
+            scope = key[: -(len(tf_leaf) + 1)]
+            return scope.replace("/", ".") + "." + pt_leaf, False
+    if key.endswith("/W"):
+        return key[:-2].replace("/", ".") + ".weight", True
+    if key.endswith("/b"):
+        return key[:-2].replace("/", ".") + ".bias", False
+    return None, False
+
+
+def torch_key_to_tf(key: str, shape) -> str | None:
+    """Inverse mapping for the saver. Returns None for non-checkpoint keys
+    (e.g. num_batches_tracked)."""
+    if key.endswith(".num_batches_tracked"):
+        return None
+    parts = key.split(".")
+    leaf = parts[-1]
+    scope = "/".join(parts[:-1])
+    # TF scope names contain a literal dot: 'downsample.0' / 'downsample.1'
+    # (model_utils.py:33-34); our Sequential child splits it into two parts.
+    scope = scope.replace("downsample/0", "downsample.0")
+    scope = scope.replace("downsample/1", "downsample.1")
+    if leaf == "running_mean":
+        return scope + "/mean/EMA"
+    if leaf == "running_var":
+        return scope + "/variance/EMA"
+    if leaf == "weight":
+        if len(shape) == 4:
+            return scope + "/W"
+        return scope + "/gamma"       # 1-D affine = norm gamma
+    if leaf == "bias":
+        # conv bias vs norm beta: disambiguated by the sibling weight's rank
+        return scope + ("/b" if _sibling_is_conv.get(key, True) else "/beta")
+    return None
+
+
+# populated by save_npz while walking the state dict
+_sibling_is_conv: Dict[str, bool] = {}
+
+
+def load_npz(model: torch.nn.Module, path: str, strict: bool = True) -> None:
+    """Load a reference-layout ``.npz`` into the model (HWIO -> OIHW)."""
+    archive = np.load(path)
+    sd = model.state_dict()
+    new_sd = {}
+    missing_src = []
+    for tf_key in archive.files:
+        pt_key, transpose = tf_key_to_torch(tf_key)
+        if pt_key is None:
+            continue
+        arr = archive[tf_key]
+        if transpose:
+            arr = np.transpose(arr, (3, 2, 0, 1))   # HWIO -> OIHW
+        if pt_key not in sd:
+            missing_src.append(tf_key)
+            continue
+        t = torch.from_numpy(np.ascontiguousarray(arr))
+        if tuple(t.shape) != tuple(sd[pt_key].shape):
+            raise ValueError(
+                f"shape mismatch for {tf_key} -> {pt_key}: "
+                f"{tuple(t.shape)} vs {tuple(sd[pt_key].shape)}")
+        new_sd[pt_key] = t.to(sd[pt_key].dtype)
+    if strict and missing_src:
+        raise KeyError(f"npz keys with no matching module param: {missing_src}")
+    not_loaded = [k for k in sd
+                  if k not in new_sd and not k.endswith("num_batches_tracked")]
+    if strict and not_loaded:
+        raise KeyError(f"model params not found in npz: {not_loaded}")
+    model.load_state_dict(new_sd, strict=False)
+
+
+def save_npz(model: torch.nn.Module, path: str) -> None:
+    """Save model weights in the reference ``.npz`` layout (OIHW -> HWIO)."""
+    sd = model.state_dict()
+    # classify bias leaves: conv bias iff the sibling weight is 4-D
+    _sibling_is_conv.clear()
+    for k, v in sd.items():
+        if k.endswith(".bias"):
+            w = sd.get(k[:-5] + ".weight")
+            _sibling_is_conv[k] = w is not None and w.dim() == 4
+    out = {}
+    for k, v in sd.items():
+        tf_key = torch_key_to_tf(k, v.shape)
+        if tf_key is None:
+            continue
+        arr = v.detach().cpu().float().numpy()
+        if tf_key.endswith("/W"):
+            arr = np.transpose(arr, (2, 3, 1, 0))   # OIHW -> HWIO
+        out[tf_key] = arr
+    np.savez(path, **out)
+
+
+def expected_npz_keys(model: torch.nn.Module) -> list:
+    """The TF key set this model would save — used by schema tests."""
+    sd = model.state_dict()
+    _sibling_is_conv.clear()
+    for k, v in sd.items():
+        if k.endswith(".bias"):
+            w = sd.get(k[:-5] + ".weight")
+            _sibling_is_conv[k] = w is not None and w.dim() == 4
+    keys = []
+    for k, v in sd.items():
+        tf_key = torch_key_to_tf(k, v.shape)
+        if tf_key is not None:
+            keys.append(tf_key)
+    return sorted(keys)

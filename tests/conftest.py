@@ -1,0 +1,27 @@
+import os
+import random
+
+import numpy as np
+import pytest
+import torch
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD GPU (run via gpurun on MI355X)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this container")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    random.seed(0)
+    np.random.seed(0)
+    torch.manual_seed(0)

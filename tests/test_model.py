@@ -1,0 +1,68 @@
+"""Model-level shape / parameter / mode tests (CPU, golden path)."""
+import pytest
+import torch
+
+from raft_amd import RAFT, RaftConfig
+
+
+def test_small_forward_config1():
+    """BASELINE config 1: raft-small, 2x128x256, 12 iters, CPU."""
+    m = RAFT(RaftConfig(small=True)).eval()
+    x1 = torch.rand(1, 3, 128, 256)
+    x2 = torch.rand(1, 3, 128, 256)
+    with torch.no_grad():
+        out = m(x1, x2, iters=12)
+    assert out.shape == (1, 2, 128, 256)
+    assert torch.isfinite(out).all()
+
+
+def test_things_forward():
+    m = RAFT(RaftConfig(small=False)).eval()
+    x1 = torch.rand(2, 3, 64, 96)
+    x2 = torch.rand(2, 3, 64, 96)
+    with torch.no_grad():
+        out = m(x1, x2, iters=3)
+    assert out.shape == (2, 2, 64, 96)
+
+
+def test_param_counts_match_official_scale():
+    # official RAFT: 5.3M (things) / 1.0M (small) — BASELINE.md
+    things = sum(p.numel() for p in RAFT(RaftConfig(small=False)).parameters())
+    small = sum(p.numel() for p in RAFT(RaftConfig(small=True)).parameters())
+    assert 5.0e6 < things < 5.5e6
+    assert 0.9e6 < small < 1.1e6
+
+
+def test_train_mode_returns_sequence():
+    m = RAFT(RaftConfig(small=True))
+    x1 = torch.rand(1, 3, 64, 96)
+    x2 = torch.rand(1, 3, 64, 96)
+    preds = m(x1, x2, iters=4, test_mode=False)
+    assert len(preds) == 4
+    for p in preds:
+        assert p.shape == (1, 2, 64, 96)
+    # gradients flow back to encoder weights through the sequence loss
+    loss = sum((p ** 2).mean() for p in preds)
+    loss.backward()
+    assert m.fnet.conv1.weight.grad is not None
+    assert m.update_block.gru.convz.weight.grad is not None
+
+
+def test_flow_init_warm_start():
+    m = RAFT(RaftConfig(small=True)).eval()
+    x1 = torch.rand(1, 3, 64, 96)
+    x2 = torch.rand(1, 3, 64, 96)
+    with torch.no_grad():
+        o0 = m(x1, x2, iters=2)
+        finit = torch.zeros(1, 2, 8, 12)
+        o1 = m(x1, x2, iters=2, flow_init=finit)
+    assert torch.allclose(o0, o1, atol=1e-5)
+
+
+def test_dynamic_shapes():
+    """Reference hardwired (432,1024); rebuild supports any /8 shape."""
+    m = RAFT(RaftConfig(small=True)).eval()
+    for h, w in [(64, 64), (96, 160), (128, 256)]:
+        with torch.no_grad():
+            out = m(torch.rand(1, 3, h, w), torch.rand(1, 3, h, w), iters=2)
+        assert out.shape == (1, 2, h, w)
