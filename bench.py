@@ -1,0 +1,189 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — BASELINE.json headline config.
+
+Measures frames/sec (whole node) for raft-things inference at 2x436x1024,
+bf16, 32 iters (BASELINE config 2) on N GPUs of one node (weak scaling: one
+frame-pair stream per rank, synthetic data, random-init weights).
+
+  python bench.py --gpus N --steps K --warmup W
+  # N>1 is launched by the driver via torch.distributed.run (one rank/GPU,
+  # RCCL over xGMI); rank 0 prints ONE JSON line.
+
+Extra modes (not used by the driver contract): --train benches the DP
+training step (config 3 shape), --small / --height/--width/--iters override
+the model/config.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=16)
+    p.add_argument("--warmup", type=int, default=4)
+    p.add_argument("--batch", type=int, default=1, help="per-rank batch")
+    p.add_argument("--height", type=int, default=436)
+    p.add_argument("--width", type=int, default=1024)
+    p.add_argument("--iters", type=int, default=32)
+    p.add_argument("--small", action="store_true")
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--train", action="store_true",
+                   help="bench the training step (config 3) instead")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture of the inference step")
+    return p.parse_args()
+
+
+def pad8(x):
+    import torch.nn.functional as F
+    h, w = x.shape[-2:]
+    ph = (-h) % 8
+    pw = (-w) % 8
+    if ph or pw:
+        x = F.pad(x, (0, pw, 0, ph))
+    return x
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    have_gpu = torch.cuda.is_available()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group(backend="nccl" if have_gpu else "gloo")
+        if have_gpu:
+            torch.cuda.set_device(local_rank)
+
+    from raft_amd import RAFT, RaftConfig
+
+    if not have_gpu:
+        # CPU fallback: plumbing config (BASELINE config 1) so the script
+        # terminates; the driver runs the real bench on an MI355X.
+        args.small, args.height, args.width = True, 128, 256
+        args.iters, args.dtype = 12, "fp32"
+
+    dev = torch.device(f"cuda:{local_rank}" if have_gpu else "cpu")
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+
+    torch.manual_seed(1234 + rank)
+    model = RAFT(RaftConfig(small=args.small)).to(dev)
+    if not args.train:
+        model = model.eval().to(dtype)
+
+    H, W = args.height, args.width
+    x1 = pad8(torch.rand(args.batch, 3, H, W)).to(dev, dtype)
+    x2 = pad8(torch.rand(args.batch, 3, H, W)).to(dev, dtype)
+
+    if args.train:
+        run_step, teardown = _make_train_step(model, args, dev, dist)
+    else:
+        run_step, teardown = _make_infer_step(model, x1, x2, args)
+
+    def sync():
+        if dist is not None:
+            dist.barrier()
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        run_step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    if dist is not None:
+        t = torch.tensor([elapsed], device=dev if have_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    frames = args.steps * args.batch * world
+    fps = frames / elapsed
+    if rank == 0:
+        model_name = "raft-small" if args.small else "raft-things"
+        mode = "training" if args.train else "inference"
+        result = {
+            "metric": f"frames/sec (whole node), {model_name} "
+                      f"2x{args.height}x{args.width}, {args.iters} iters, "
+                      f"{mode}",
+            "value": fps,
+            "unit": "frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": 1e3 * elapsed / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,   # reference publishes no numbers
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": model_name,
+                "global_batch": args.batch * world,
+                "height": args.height,
+                "width": args.width,
+                "iters": args.iters,
+                "parallelism": f"dp{world}",
+                "device": "cpu-fallback" if not have_gpu else "mi355x",
+            },
+        }
+        print(json.dumps(result), flush=True)
+    teardown()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+def _make_infer_step(model, x1, x2, args):
+    iters = args.iters
+
+    def step():
+        with torch.no_grad():
+            out = model(x1, x2, iters=iters)
+        return out
+
+    return step, lambda: None
+
+
+def _make_train_step(model, args, dev, dist):
+    """Config 3: training step, sequence loss, AdamW, DP grad all-reduce."""
+    from raft_amd.engine.trainer import sequence_loss
+    if dist is not None:
+        from raft_amd.parallel.ddp import BucketedDDP
+        model = BucketedDDP(model)
+    opt = torch.optim.AdamW(model.parameters(), lr=4e-4, weight_decay=1e-5)
+    H = args.height if args.height != 436 else 368
+    W = args.width if args.width != 1024 else 768
+    b = args.batch if args.batch > 1 else 2
+    x1 = torch.rand(b, 3, H, W, device=dev)
+    x2 = torch.rand(b, 3, H, W, device=dev)
+    gt = torch.randn(b, 2, H, W, device=dev)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        preds = model(x1, x2, iters=args.iters if args.iters != 32 else 12,
+                      test_mode=False)
+        loss = sequence_loss(preds, gt)
+        loss.backward()
+        if dist is not None:
+            model.finish_gradient_sync()
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
+        opt.step()
+        return loss
+
+    return step, lambda: None
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,108 @@
+"""Training engine: RAFT sequence loss + AdamW/one-cycle + DP.
+
+Designed from the RAFT paper recipe (the reference has NO training path —
+its build_graph returns literal 0.0, networks/RAFT.py:141, and the train
+mode has no body):
+  * sequence loss  L = sum_i gamma^(N-1-i) * |f_gt - f_i|_1, gamma = 0.8,
+  * AdamW + one-cycle LR, gradient clipping at 1.0,
+  * DP via raft_amd.parallel.BucketedDDP (RCCL over xGMI).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+MAX_FLOW = 400.0
+
+
+def sequence_loss(flow_preds: List[torch.Tensor], flow_gt: torch.Tensor,
+                  gamma: float = 0.8,
+                  valid: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Exponentially weighted L1 over the iteration sequence (RAFT paper
+    eq. 7). Pixels with |gt| >= MAX_FLOW are excluded."""
+    n = len(flow_preds)
+    mag = torch.sum(flow_gt ** 2, dim=1).sqrt()
+    v = (mag < MAX_FLOW)
+    if valid is not None:
+        v = v & (valid >= 0.5)
+    v = v.float().unsqueeze(1)
+    loss = flow_preds[0].new_zeros(())
+    for i, pred in enumerate(flow_preds):
+        w = gamma ** (n - i - 1)
+        loss = loss + w * (v * (pred.float() - flow_gt).abs()).mean()
+    return loss
+
+
+def epe(flow_pred: torch.Tensor, flow_gt: torch.Tensor) -> torch.Tensor:
+    """End-point error (the north-star quality metric; the reference never
+    implemented it — SURVEY.md §5.5)."""
+    return torch.norm(flow_pred - flow_gt, p=2, dim=1).mean()
+
+
+@dataclass
+class TrainConfig:
+    lr: float = 4e-4
+    weight_decay: float = 1e-5        # reference RAFT.py:14
+    epsilon: float = 1e-8
+    num_steps: int = 100
+    iters: int = 12
+    gamma: float = 0.8
+    clip: float = 1.0
+    batch: int = 2
+    height: int = 368
+    width: int = 768
+
+
+class Trainer:
+    """Single-node trainer; wraps the model in BucketedDDP when distributed
+    is initialized (one process per GPU over RCCL)."""
+
+    def __init__(self, model: nn.Module, cfg: TrainConfig,
+                 device: Optional[torch.device] = None):
+        self.cfg = cfg
+        self.device = device or (
+            torch.device("cuda") if torch.cuda.is_available()
+            else torch.device("cpu"))
+        model = model.to(self.device)
+        self.raw_model = model
+        if dist.is_available() and dist.is_initialized() and \
+                dist.get_world_size() > 1:
+            from raft_amd.parallel.ddp import BucketedDDP
+            self.model = BucketedDDP(model)
+            self.distributed = True
+        else:
+            self.model = model
+            self.distributed = False
+        self.optimizer = torch.optim.AdamW(
+            model.parameters(), lr=cfg.lr, weight_decay=cfg.weight_decay,
+            eps=cfg.epsilon)
+        self.scheduler = torch.optim.lr_scheduler.OneCycleLR(
+            self.optimizer, max_lr=cfg.lr,
+            total_steps=cfg.num_steps + 10, pct_start=0.05,
+            cycle_momentum=False, anneal_strategy="linear")
+        self.step_count = 0
+
+    def step(self, image1: torch.Tensor, image2: torch.Tensor,
+             flow_gt: torch.Tensor,
+             valid: Optional[torch.Tensor] = None) -> dict:
+        self.optimizer.zero_grad(set_to_none=True)
+        preds = self.model(image1, image2, iters=self.cfg.iters,
+                           test_mode=False)
+        loss = sequence_loss(preds, flow_gt, self.cfg.gamma, valid)
+        loss.backward()
+        if self.distributed:
+            self.model.finish_gradient_sync()
+        torch.nn.utils.clip_grad_norm_(self.raw_model.parameters(),
+                                       self.cfg.clip)
+        self.optimizer.step()
+        if self.step_count + 1 < self.scheduler.total_steps:
+            self.scheduler.step()
+        self.step_count += 1
+        with torch.no_grad():
+            e = epe(preds[-1].float(), flow_gt)
+        return {"loss": float(loss.detach()), "epe": float(e),
+                "lr": self.optimizer.param_groups[0]["lr"]}

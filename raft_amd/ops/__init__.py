@@ -17,8 +17,10 @@ _hip_import_error: Exception | None = None
 
 if os.environ.get("RAFT_AMD_FORCE_TORCH", "0") != "1":
     try:
-        from raft_amd.ops import _hip_ops as _hip_mod  # type: ignore
-        _hip_ops = _hip_mod
+        import importlib
+        # NOTE: must be importlib (not `from raft_amd.ops import _hip_ops`)
+        # — the `_hip_ops = None` sentinel above would shadow the submodule.
+        _hip_ops = importlib.import_module("raft_amd.ops._hip_ops")
     except ImportError as e:  # pragma: no cover - exercised on GPU boxes
         _hip_import_error = e
 

@@ -1,0 +1,208 @@
+// K4 (pointwise part): fused ConvGRU gate math h' = (1-sigma(z))h + sigma(z)tanh(q)
+// with analytic backward (model_utils.py:146,154,168), and
+// K5: 8x convex upsample with the 9-tap softmax fused in, fwd + bwd
+// (networks/RAFT.py:119-134).
+
+#include "common.h"
+
+RAFT_DEV float sigmoidf(float x) { return 1.0f / (1.0f + __expf(-x)); }
+
+// ---------------------------------------------------------------- GRU gates
+extern "C" __global__ void gru_gates_fwd_f32(
+    const float* __restrict__ h, const float* __restrict__ z_act,
+    const float* __restrict__ q_act, float* __restrict__ out,
+    long long n) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        const float z = sigmoidf(z_act[i]);
+        const float q = tanhf(q_act[i]);
+        out[i] = (1.0f - z) * h[i] + z * q;
+    }
+}
+
+extern "C" __global__ void gru_gates_bwd_f32(
+    const float* __restrict__ go, const float* __restrict__ h,
+    const float* __restrict__ z_act, const float* __restrict__ q_act,
+    float* __restrict__ gh, float* __restrict__ gz, float* __restrict__ gq,
+    long long n) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        const float g = go[i];
+        const float z = sigmoidf(z_act[i]);
+        const float q = tanhf(q_act[i]);
+        gh[i] = g * (1.0f - z);
+        gz[i] = g * (q - h[i]) * z * (1.0f - z);
+        gq[i] = g * z * (1.0f - q * q);
+    }
+}
+
+// ------------------------------------------------------------ convex upsample
+// One 64-lane wave per 1/8-res cell (y, x); lane s -> subpixel
+// (dy = s/8, dx = s%8). mask channel c = k*64 + s (TF NHWC 576 reshaped to
+// (9,1,8,8), RAFT.py:125); softmax over the 9 taps; taps are the 3x3
+// zero-padded neighborhood of 8*flow. Block = 256 threads = 4 cells.
+extern "C" __global__ __launch_bounds__(256)
+void convex_upsample_fwd_f32(
+    const float* __restrict__ flow,   // [B, 2, H, W]
+    const float* __restrict__ mask,   // [B, 576, H, W]
+    float* __restrict__ out,          // [B, 2, 8H, 8W]
+    int B, int H, int W) {
+    const long long cell = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (cell >= (long long)B * H * W) return;
+    const int s = threadIdx.x & 63;
+    const int x = (int)(cell % W);
+    const int y = (int)((cell / W) % H);
+    const int b = (int)(cell / ((long long)W * H));
+    const long long HW = (long long)H * W;
+    const long long base = (long long)b * 576 * HW + (long long)y * W + x;
+
+    // per-lane 3x3 flow neighborhood (zero-padded SAME), scaled x8
+    float f0[9], f1[9];
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        const int ny = y + k / 3 - 1;
+        const int nx = x + k % 3 - 1;
+        const bool ok = (ny >= 0 && ny < H && nx >= 0 && nx < W);
+        const long long fi = ((long long)b * 2 * H + ny) * W + nx;
+        f0[k] = ok ? 8.0f * flow[fi] : 0.0f;
+        f1[k] = ok ? 8.0f * flow[fi + HW] : 0.0f;
+    }
+
+    float m[9], mx = -1e30f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        m[k] = mask[base + (long long)(k * 64 + s) * HW];
+        mx = fmaxf(mx, m[k]);
+    }
+    float denom = 0.f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        m[k] = __expf(m[k] - mx);
+        denom += m[k];
+    }
+    float o0 = 0.f, o1 = 0.f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        const float p = m[k] / denom;
+        o0 += p * f0[k];
+        o1 += p * f1[k];
+    }
+    const int oy = 8 * y + s / 8;
+    const int ox = 8 * x + s % 8;
+    const long long oi = (((long long)b * 2) * 8 * H + oy) * 8 * W + ox;
+    out[oi] = o0;
+    out[oi + 64 * HW] = o1;
+}
+
+extern "C" __global__ __launch_bounds__(256)
+void convex_upsample_bwd_f32(
+    const float* __restrict__ grad_up,  // [B, 2, 8H, 8W]
+    const float* __restrict__ flow, const float* __restrict__ mask,
+    float* __restrict__ grad_flow,      // [B, 2, H, W] (pre-zeroed)
+    float* __restrict__ grad_mask,      // [B, 576, H, W]
+    int B, int H, int W) {
+    const long long cell = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (cell >= (long long)B * H * W) return;
+    const int s = threadIdx.x & 63;
+    const int x = (int)(cell % W);
+    const int y = (int)((cell / W) % H);
+    const int b = (int)(cell / ((long long)W * H));
+    const long long HW = (long long)H * W;
+    const long long base = (long long)b * 576 * HW + (long long)y * W + x;
+
+    float f0[9], f1[9];
+    bool ok[9];
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        const int ny = y + k / 3 - 1;
+        const int nx = x + k % 3 - 1;
+        ok[k] = (ny >= 0 && ny < H && nx >= 0 && nx < W);
+        const long long fi = ((long long)b * 2 * H + ny) * W + nx;
+        f0[k] = ok[k] ? 8.0f * flow[fi] : 0.0f;
+        f1[k] = ok[k] ? 8.0f * flow[fi + HW] : 0.0f;
+    }
+    float m[9], mx = -1e30f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        m[k] = mask[base + (long long)(k * 64 + s) * HW];
+        mx = fmaxf(mx, m[k]);
+    }
+    float denom = 0.f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) { m[k] = __expf(m[k] - mx); denom += m[k]; }
+
+    const int oy = 8 * y + s / 8;
+    const int ox = 8 * x + s % 8;
+    const long long oi = (((long long)b * 2) * 8 * H + oy) * 8 * W + ox;
+    const float gu0 = grad_up[oi];
+    const float gu1 = grad_up[oi + 64 * HW];
+
+    float p[9], gk[9], S = 0.f;
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        p[k] = m[k] / denom;
+        gk[k] = f0[k] * gu0 + f1[k] * gu1;
+        S += p[k] * gk[k];
+    }
+#pragma unroll
+    for (int k = 0; k < 9; ++k)
+        grad_mask[base + (long long)(k * 64 + s) * HW] = p[k] * (gk[k] - S);
+
+    // flow grad: wave-reduce the 64 subpixels' contribution per neighbor,
+    // one atomicAdd per (k, channel) from lane 0.
+#pragma unroll
+    for (int k = 0; k < 9; ++k) {
+        float v0 = p[k] * gu0 * 8.0f;
+        float v1 = p[k] * gu1 * 8.0f;
+        for (int off = 32; off > 0; off >>= 1) {
+            v0 += __shfl_down(v0, off, 64);
+            v1 += __shfl_down(v1, off, 64);
+        }
+        if (s == 0 && ok[k]) {
+            const int ny = y + k / 3 - 1;
+            const int nx = x + k % 3 - 1;
+            const long long fi = ((long long)b * 2 * H + ny) * W + nx;
+            atomicAdd(&grad_flow[fi], v0);
+            atomicAdd(&grad_flow[fi + HW], v1);
+        }
+    }
+}
+
+// ----------------------------------------------------------- host launchers
+extern "C" void launch_gru_gates_fwd_f32(const float* h, const float* z,
+                                         const float* q, float* out,
+                                         long long n, hipStream_t s) {
+    int blocks = (int)min((n + 255) / 256, (long long)2048);
+    hipLaunchKernelGGL(gru_gates_fwd_f32, dim3(blocks), dim3(256), 0, s,
+                       h, z, q, out, n);
+}
+
+extern "C" void launch_gru_gates_bwd_f32(const float* go, const float* h,
+                                         const float* z, const float* q,
+                                         float* gh, float* gz, float* gq,
+                                         long long n, hipStream_t s) {
+    int blocks = (int)min((n + 255) / 256, (long long)2048);
+    hipLaunchKernelGGL(gru_gates_bwd_f32, dim3(blocks), dim3(256), 0, s,
+                       go, h, z, q, gh, gz, gq, n);
+}
+
+extern "C" void launch_convex_upsample_fwd_f32(const float* flow,
+                                               const float* mask, float* out,
+                                               int B, int H, int W,
+                                               hipStream_t s) {
+    long long cells = (long long)B * H * W;
+    hipLaunchKernelGGL(convex_upsample_fwd_f32,
+                       dim3((unsigned)((cells + 3) / 4)), dim3(256), 0, s,
+                       flow, mask, out, B, H, W);
+}
+
+extern "C" void launch_convex_upsample_bwd_f32(
+    const float* grad_up, const float* flow, const float* mask,
+    float* grad_flow, float* grad_mask, int B, int H, int W, hipStream_t s) {
+    long long cells = (long long)B * H * W;
+    hipLaunchKernelGGL(convex_upsample_bwd_f32,
+                       dim3((unsigned)((cells + 3) / 4)), dim3(256), 0, s,
+                       grad_up, flow, mask, grad_flow, grad_mask, B, H, W);
+}

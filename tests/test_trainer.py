@@ -1,0 +1,43 @@
+"""Single-process training engine tests."""
+import torch
+
+from raft_amd import RAFT, RaftConfig
+from raft_amd.engine.trainer import Trainer, TrainConfig, epe, sequence_loss
+
+
+def test_sequence_loss_weighting():
+    gt = torch.zeros(1, 2, 8, 8)
+    p1 = torch.ones(1, 2, 8, 8)
+    # single prediction: weight gamma^0 = 1 -> mean |1| = 1
+    assert abs(float(sequence_loss([p1], gt)) - 1.0) < 1e-6
+    # two predictions: gamma^1 * L(p0) + gamma^0 * L(p1)
+    l = float(sequence_loss([p1, 2 * p1], gt, gamma=0.8))
+    assert abs(l - (0.8 * 1.0 + 1.0 * 2.0)) < 1e-5
+
+
+def test_sequence_loss_excludes_large_flow():
+    gt = torch.zeros(1, 2, 4, 4)
+    gt[0, 0, 0, 0] = 500.0   # beyond MAX_FLOW -> excluded
+    pred = torch.zeros(1, 2, 4, 4)
+    assert float(sequence_loss([pred], gt)) == 0.0
+
+
+def test_epe():
+    a = torch.zeros(1, 2, 4, 4)
+    b = torch.zeros(1, 2, 4, 4)
+    b[:, 0] = 3.0
+    b[:, 1] = 4.0
+    assert abs(float(epe(a, b)) - 5.0) < 1e-6
+
+
+def test_trainer_loss_decreases_on_fixed_batch():
+    torch.manual_seed(0)
+    cfg = TrainConfig(num_steps=8, iters=2, lr=1e-3,
+                      height=64, width=96, batch=1)
+    tr = Trainer(RAFT(RaftConfig(small=True)), cfg,
+                 device=torch.device("cpu"))
+    x1 = torch.rand(1, 3, 64, 96)
+    x2 = torch.rand(1, 3, 64, 96)
+    gt = torch.zeros(1, 2, 64, 96)
+    losses = [tr.step(x1, x2, gt)["loss"] for _ in range(8)]
+    assert losses[-1] < losses[0], losses
