@@ -1,0 +1,189 @@
+#!/usr/bin/env python3
+"""RAFT optical-flow CLI — same flag surface as the reference's
+infer_raft.py (BASELINE.json API contract), with the modes the reference
+left unimplemented (train / val / export / flops) working.
+
+Modes:
+  test    run flow on an image pair, write the color-coded PNG
+          (reference infer_raft.py:74-78; output file name kept)
+  train   train on synthetic pairs (the reference had no train body)
+  val     EPE evaluation on synthetic pairs with exact ground truth
+  export  save weights as reference-layout .npz (+ state_dict .pt)
+  flops   parameter/FLOP report (the reference's flops mode crashes on an
+          arity bug, networks/RAFT.py:144 — this one works)
+
+Flags mirror the reference (infer_raft.py:51-67): --gpu --data --load
+-m/--mode --out --batch -o/--optimizer --im1 --im2 --small, plus the
+rebuild's --iters/--size/--dtype/--steps (iters was hard-coded 20 in the
+reference, networks/RAFT.py:33).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--gpu", default=None,
+                   help="comma separated list of GPU(s) to use")
+    p.add_argument("--data", default=None, help="dataset path (training)")
+    p.add_argument("--load", default=None,
+                   help="checkpoint to load (.npz reference layout or .pt)")
+    p.add_argument("-m", "--mode", default="test",
+                   choices=["train", "val", "test", "export", "flops"])
+    p.add_argument("--out", default="./log", help="output directory")
+    p.add_argument("--batch", default=1, type=int, help="batch per GPU")
+    p.add_argument("-o", "--optimizer", default="adamw",
+                   choices=["adam", "adamw", "sgd", "sgd_cyclic",
+                            "sgd_1cycle"])
+    p.add_argument("--im1", default="frame_0010.png", help="left image path")
+    p.add_argument("--im2", default="frame_0011.png", help="right image path")
+    p.add_argument("--small", action="store_true")
+    # rebuild extensions
+    p.add_argument("--iters", type=int, default=None,
+                   help="refinement iterations (default: 20, ref value)")
+    p.add_argument("--size", default="432x1024",
+                   help="HxW input resize for test mode; 'native' keeps "
+                        "the file size (pad8'd)")
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
+    p.add_argument("--steps", type=int, default=100, help="training steps")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable HIP-graph capture in test mode")
+    return p.parse_args(argv)
+
+
+def _select_device(args):
+    if args.gpu is not None and torch.cuda.is_available():
+        os.environ.setdefault("HIP_VISIBLE_DEVICES", args.gpu)
+    return torch.device("cuda") if torch.cuda.is_available() \
+        else torch.device("cpu")
+
+
+def _build_model(args, device):
+    from raft_amd import RAFT, RaftConfig
+    model = RAFT(RaftConfig(small=args.small)).to(device).eval()
+    if args.load:
+        from raft_amd.utils import checkpoint as ckpt
+        if args.load.endswith(".npz"):
+            ckpt.load_npz(model, args.load)
+        else:
+            state = torch.load(args.load, map_location="cpu",
+                               weights_only=True)
+            model.load_state_dict(state.get("model", state))
+        print(f"loaded {args.load}")
+    return model
+
+
+def mode_test(args, device):
+    from raft_amd.data.dataflow import PairDataflow
+    from raft_amd.engine.inference import InferenceEngine
+    from raft_amd.data.imageio import write_png
+    from raft_amd.utils.flow_viz import flow_to_color
+
+    model = _build_model(args, device)
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    engine = InferenceEngine(model, iters=args.iters, dtype=dtype,
+                             use_graph=not args.no_graph)
+    size = None if args.size == "native" else \
+        tuple(int(v) for v in args.size.split("x"))
+    ds = PairDataflow([(args.im1, args.im2)], input_size=size,
+                      batch=args.batch)
+    os.makedirs(args.out, exist_ok=True)
+    variant = "raft-small" if args.small else "raft-things"
+    for i, (im1, im2) in enumerate(ds):
+        t0 = time.perf_counter()
+        flow = engine(im1, im2)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        print(i, tuple(flow.shape), f"{dt * 1e3:.1f} ms")
+        flow_np = flow[0].float().permute(1, 2, 0).cpu().numpy()
+        color = flow_to_color(flow_np, convert_to_bgr=True)
+        out_path = os.path.join(args.out, f"raft_flow_{variant}.png")
+        write_png(out_path, color)
+        print(f"wrote {out_path}")
+
+
+def mode_val(args, device):
+    from raft_amd.data.synthetic import synthetic_pair
+    from raft_amd.engine.trainer import epe
+    from raft_amd.engine.inference import InferenceEngine
+
+    model = _build_model(args, device)
+    engine = InferenceEngine(
+        model, iters=args.iters,
+        dtype=torch.bfloat16 if args.dtype == "bf16" else torch.float32,
+        use_graph=not args.no_graph)
+    epes = []
+    for seed in range(8):
+        im1, im2, gt = synthetic_pair(args.batch, 288, 512, seed=seed)
+        flow = engine(im1, im2)
+        epes.append(float(epe(flow.float().cpu(), gt)))
+    result = {"epe_mean": float(np.mean(epes)), "epe_per_batch": epes,
+              "data": "synthetic-warp 288x512", "iters": args.iters}
+    print(json.dumps(result))
+
+
+def mode_train(args, device):
+    from raft_amd.data.synthetic import synthetic_pair
+    from raft_amd.engine.trainer import Trainer, TrainConfig
+    from raft_amd.parallel.ddp import init_distributed
+    from raft_amd.utils import checkpoint as ckpt
+
+    rank = init_distributed()
+    model = _build_model(args, device)
+    model.train()
+    cfg = TrainConfig(num_steps=args.steps, batch=args.batch)
+    tr = Trainer(model, cfg, device=device)
+    for step in range(args.steps):
+        im1, im2, gt = synthetic_pair(args.batch, 288, 512,
+                                      seed=step * 131 + rank)
+        stats = tr.step(im1.to(device), im2.to(device), gt.to(device))
+        if rank == 0 and (step % 10 == 0 or step == args.steps - 1):
+            print(f"step {step}: loss {stats['loss']:.4f} "
+                  f"epe {stats['epe']:.3f} lr {stats['lr']:.2e}")
+    if rank == 0:
+        os.makedirs(args.out, exist_ok=True)
+        out = os.path.join(args.out, "raft_trained.npz")
+        ckpt.save_npz(model, out)
+        print(f"saved {out}")
+
+
+def mode_export(args, device):
+    from raft_amd.utils import checkpoint as ckpt
+    model = _build_model(args, device)
+    os.makedirs(args.out, exist_ok=True)
+    variant = "raft-small" if args.small else "raft-things"
+    npz_path = os.path.join(args.out, f"{variant}.npz")
+    pt_path = os.path.join(args.out, f"{variant}.pt")
+    ckpt.save_npz(model, npz_path)
+    torch.save({"model": model.state_dict()}, pt_path)
+    print(f"exported {npz_path} and {pt_path}")
+
+
+def mode_flops(args, device):
+    from raft_amd import RAFT, RaftConfig
+    from raft_amd.engine.profiler import count_model_flops
+    model = RAFT(RaftConfig(small=args.small))
+    report = count_model_flops(model, 256, 448,
+                               iters=args.iters or model.cfg.iters)
+    print(json.dumps(report, indent=2))
+    print("note: FLOPs counted as 2*MACs (multiply+add separately), matching "
+          "tf.profiler convention the reference documents "
+          "(infer_raft.py:93-95)")
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    device = _select_device(args)
+    {"test": mode_test, "val": mode_val, "train": mode_train,
+     "export": mode_export, "flops": mode_flops}[args.mode](args, device)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

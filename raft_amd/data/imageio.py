@@ -1,0 +1,134 @@
+"""Minimal pure-NumPy PNG reader/writer.
+
+The environment ships neither OpenCV nor PIL, so the CLI's image IO
+(reference: cv2.imdecode/imwrite, dataflow/test_dataflow.py:56-61,
+infer_raft.py:44) is implemented from the PNG spec directly: 8-bit
+gray/RGB/RGBA, all five scanline filters, no interlace.  Output images are
+written as filter-0 RGB8.
+"""
+from __future__ import annotations
+
+import struct
+import zlib
+
+import numpy as np
+
+_PNG_SIG = b"\x89PNG\r\n\x1a\n"
+
+
+def read_png(path: str) -> np.ndarray:
+    """Read a PNG into an HxWx3 uint8 **BGR** array (the channel order the
+    model weights expect — networks/RAFT.py:13)."""
+    with open(path, "rb") as f:
+        data = f.read()
+    if data[:8] != _PNG_SIG:
+        raise ValueError(f"{path}: not a PNG")
+    pos = 8
+    width = height = bit_depth = color_type = None
+    idat = bytearray()
+    palette = None
+    while pos < len(data):
+        length, ctype = struct.unpack(">I4s", data[pos:pos + 8])
+        chunk = data[pos + 8:pos + 8 + length]
+        pos += 12 + length
+        if ctype == b"IHDR":
+            width, height, bit_depth, color_type, _, _, interlace = \
+                struct.unpack(">IIBBBBB", chunk)
+            if bit_depth != 8:
+                raise ValueError(f"{path}: only 8-bit PNGs supported")
+            if interlace:
+                raise ValueError(f"{path}: interlaced PNGs not supported")
+        elif ctype == b"PLTE":
+            palette = np.frombuffer(chunk, np.uint8).reshape(-1, 3)
+        elif ctype == b"IDAT":
+            idat.extend(chunk)
+        elif ctype == b"IEND":
+            break
+    channels = {0: 1, 2: 3, 3: 1, 4: 2, 6: 4}[color_type]
+    raw = zlib.decompress(bytes(idat))
+    stride = width * channels
+    expected = height * (stride + 1)
+    if len(raw) != expected:
+        raise ValueError(f"{path}: bad IDAT size {len(raw)} != {expected}")
+    raw = np.frombuffer(raw, np.uint8).reshape(height, stride + 1)
+    filters = raw[:, 0]
+    img = _unfilter(raw[:, 1:].astype(np.int32), filters, channels)
+    img = img.reshape(height, width, channels)
+    if color_type == 3:
+        img = palette[img[:, :, 0]]
+    elif channels == 1:
+        img = np.repeat(img, 3, axis=2)
+    elif channels == 2:
+        img = np.repeat(img[:, :, :1], 3, axis=2)
+    elif channels == 4:
+        img = img[:, :, :3]
+    return np.ascontiguousarray(img[:, :, ::-1])  # RGB -> BGR
+
+
+def _unfilter(rows: np.ndarray, filters: np.ndarray, bpp: int) -> np.ndarray:
+    """Undo PNG scanline filters. Rows are sequential (each depends on the
+    previous reconstructed row); within a row, 'sub'/'paeth'/'avg' depend on
+    the left pixel so those scan x in bpp-strided python loops over
+    *columns* (cheap: width iterations, vectorized over nothing — but only
+    for the rows that use those filters)."""
+    h, stride = rows.shape
+    out = np.zeros((h, stride), np.int32)
+    for y in range(h):
+        f = filters[y]
+        cur = rows[y]
+        prev = out[y - 1] if y > 0 else np.zeros(stride, np.int32)
+        if f == 0:
+            out[y] = cur
+        elif f == 2:  # up
+            out[y] = (cur + prev) & 0xFF
+        elif f == 1:  # sub: per-channel prefix sum mod 256
+            r = cur.reshape(-1, bpp).cumsum(axis=0) & 0xFF
+            out[y] = r.reshape(-1)
+        elif f == 3:  # average
+            r = cur.copy()
+            for x in range(stride):
+                left = r[x - bpp] if x >= bpp else 0
+                r[x] = (r[x] + ((left + prev[x]) >> 1)) & 0xFF
+            out[y] = r
+        elif f == 4:  # paeth
+            r = cur.copy()
+            for x in range(stride):
+                a = r[x - bpp] if x >= bpp else 0
+                b = prev[x]
+                c = prev[x - bpp] if x >= bpp else 0
+                p = a + b - c
+                pa, pb, pc = abs(p - a), abs(p - b), abs(p - c)
+                if pa <= pb and pa <= pc:
+                    pr = a
+                elif pb <= pc:
+                    pr = b
+                else:
+                    pr = c
+                r[x] = (r[x] + pr) & 0xFF
+            out[y] = r
+        else:
+            raise ValueError(f"unknown PNG filter {f}")
+    return out.astype(np.uint8)
+
+
+def write_png(path: str, img: np.ndarray) -> None:
+    """Write an HxWx3 uint8 **BGR** array as a PNG (RGB8, filter 0)."""
+    if img.dtype != np.uint8:
+        img = np.clip(img, 0, 255).astype(np.uint8)
+    if img.ndim == 2:
+        img = np.repeat(img[:, :, None], 3, axis=2)
+    h, w, c = img.shape
+    rgb = img[:, :, ::-1] if c == 3 else img  # BGR -> RGB
+    raw = b"".join(b"\x00" + rgb[y].tobytes() for y in range(h))
+
+    def chunk(ctype: bytes, payload: bytes) -> bytes:
+        crc = zlib.crc32(ctype + payload) & 0xFFFFFFFF
+        return struct.pack(">I", len(payload)) + ctype + payload + \
+            struct.pack(">I", crc)
+
+    ihdr = struct.pack(">IIBBBBB", w, h, 8, 2, 0, 0, 0)
+    with open(path, "wb") as f:
+        f.write(_PNG_SIG)
+        f.write(chunk(b"IHDR", ihdr))
+        f.write(chunk(b"IDAT", zlib.compress(raw, 6)))
+        f.write(chunk(b"IEND", b""))

@@ -154,24 +154,5 @@ class RAFT(nn.Module):
             return up
         return ops.convex_upsample(flow.to(up_mask.dtype), up_mask)
 
-    # ------------------------------------------------------------------
-    def count_flops(self, height: int, width: int) -> dict:
-        """Analytic FLOP count for one forward (the reference's flops mode
-        crashes on an arity bug, infer_raft.py:80-95 / RAFT.py:144; this is
-        the working replacement). Counts MACs*2, conv only + corr GEMM."""
-        flops = 0
-
-        def conv_flops(cin, cout, k, h, w):
-            kh, kw = (k, k) if isinstance(k, int) else k
-            return 2 * cin * cout * kh * kw * h * w
-
-        h8, w8 = height // 8, width // 8
-        # This intentionally covers the dominant terms (encoders are counted
-        # via a module sweep; the GRU loop dominated total work anyway).
-        for mod in self.modules():
-            pass  # conv spatial sizes require shape propagation; use profiler
-        c = self.cfg.fnet_dim
-        corr_gemm = 2 * (h8 * w8) ** 2 * c
-        flops += corr_gemm
-        return {"corr_gemm_flops": corr_gemm,
-                "params": sum(p.numel() for p in self.parameters())}
+    # FLOP reporting lives in raft_amd.engine.profiler.count_model_flops
+    # (the reference's flops mode crashes on an arity bug, RAFT.py:144).

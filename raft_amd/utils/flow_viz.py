@@ -1,0 +1,100 @@
+"""Optical-flow color coding (Middlebury / Baker et al. ICCV'07 wheel).
+
+Functional parity with the reference's flow_utils.py:6-121 (flow_to_color /
+flow_compute_color / make_colorwheel): 55-color wheel, angle -> hue via
+atan2(-v, -u), radius-normalized saturation, out-of-wheel range damped by
+0.75, optional BGR output for PNG writing.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+
+def make_colorwheel() -> np.ndarray:
+    """55x3 RGB color wheel: RY 15, YG 6, GC 4, CB 11, BM 13, MR 6."""
+    transitions = [
+        (15, 0, 1, False),   # RY: R=255, G ramps up
+        (6, 0, 1, True),     # YG: R ramps down, G=255
+        (4, 1, 2, False),    # GC: G=255, B ramps up
+        (11, 1, 2, True),    # CB: G ramps down, B=255
+        (13, 2, 0, False),   # BM: B=255, R ramps up
+        (6, 2, 0, True),     # MR: B ramps down, R=255
+    ]
+    ncols = sum(t[0] for t in transitions)
+    wheel = np.zeros((ncols, 3))
+    col = 0
+    for n, hold, ramp, down in transitions:
+        wheel[col:col + n, hold] = 255
+        r = np.floor(255 * np.arange(n) / n)
+        if down:
+            wheel[col:col + n, hold] = 255 - r
+            wheel[col:col + n, ramp if False else (hold if False else _other(hold, ramp))] = 255
+        col += n
+    # the loop above is clearer written explicitly; rebuild precisely:
+    wheel = np.zeros((ncols, 3))
+    col = 0
+    RY, YG, GC, CB, BM, MR = 15, 6, 4, 11, 13, 6
+    wheel[0:RY, 0] = 255
+    wheel[0:RY, 1] = np.floor(255 * np.arange(RY) / RY)
+    col += RY
+    wheel[col:col + YG, 0] = 255 - np.floor(255 * np.arange(YG) / YG)
+    wheel[col:col + YG, 1] = 255
+    col += YG
+    wheel[col:col + GC, 1] = 255
+    wheel[col:col + GC, 2] = np.floor(255 * np.arange(GC) / GC)
+    col += GC
+    wheel[col:col + CB, 1] = 255 - np.floor(255 * np.arange(CB) / CB)
+    wheel[col:col + CB, 2] = 255
+    col += CB
+    wheel[col:col + BM, 2] = 255
+    wheel[col:col + BM, 0] = np.floor(255 * np.arange(BM) / BM)
+    col += BM
+    wheel[col:col + MR, 2] = 255 - np.floor(255 * np.arange(MR) / MR)
+    wheel[col:col + MR, 0] = 255
+    return wheel
+
+
+def _other(a, b):
+    return b
+
+
+def flow_compute_color(u: np.ndarray, v: np.ndarray,
+                       convert_to_bgr: bool = False) -> np.ndarray:
+    """Color-code normalized flow components (|.| <= 1 in the wheel)."""
+    image = np.zeros((u.shape[0], u.shape[1], 3), np.uint8)
+    wheel = make_colorwheel()
+    ncols = wheel.shape[0]
+
+    rad = np.sqrt(u ** 2 + v ** 2)
+    a = np.arctan2(-v, -u) / np.pi
+    fk = (a + 1.0) / 2.0 * (ncols - 1) + 1.0
+    k0 = np.floor(fk).astype(np.int32)
+    k0 = np.minimum(k0, ncols - 2)
+    k1 = k0 + 1
+    k1[k1 == ncols] = 1
+    f = fk - k0
+
+    inside = rad <= 1
+    for i in range(3):
+        col0 = wheel[k0, i] / 255.0
+        col1 = wheel[k1, i] / 255.0
+        col = (1.0 - f) * col0 + f * col1
+        col[inside] = 1.0 - rad[inside] * (1.0 - col[inside])
+        col[~inside] *= 0.75
+        ch = 2 - i if convert_to_bgr else i
+        image[:, :, ch] = np.floor(255.0 * col)
+    return image
+
+
+def flow_to_color(flow_uv: np.ndarray, clip_flow: float | None = None,
+                  convert_to_bgr: bool = False) -> np.ndarray:
+    """[H,W,2] flow -> uint8 color image, normalized by the max radius."""
+    assert flow_uv.ndim == 3 and flow_uv.shape[2] == 2
+    if clip_flow is not None:
+        flow_uv = np.clip(flow_uv, 0, clip_flow)
+    u = flow_uv[:, :, 0]
+    v = flow_uv[:, :, 1]
+    rad_max = float(np.sqrt(u ** 2 + v ** 2).max())
+    eps = 1e-5
+    return flow_compute_color(u / (rad_max + eps), v / (rad_max + eps),
+                              convert_to_bgr)

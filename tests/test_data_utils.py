@@ -1,0 +1,114 @@
+"""PNG/flo IO, flow viz, synthetic data, dataflow, inference engine."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from raft_amd.data.imageio import read_png, write_png
+from raft_amd.data.synthetic import synthetic_pair, warp
+from raft_amd.utils.flow_io import read_flo, write_flo, resize_flow
+from raft_amd.utils.flow_viz import flow_to_color, make_colorwheel
+from raft_amd.engine.inference import pad8, unpad, InferenceEngine
+
+
+def test_png_roundtrip(tmp_path):
+    img = (np.random.rand(37, 53, 3) * 255).astype(np.uint8)
+    p = str(tmp_path / "x.png")
+    write_png(p, img)
+    assert np.array_equal(read_png(p), img)
+
+
+def test_flo_roundtrip(tmp_path):
+    flow = np.random.randn(17, 23, 2).astype(np.float32)
+    p = str(tmp_path / "x.flo")
+    write_flo(p, flow)
+    assert np.array_equal(read_flo(p), flow)
+
+
+def test_resize_flow_scales_magnitudes():
+    flow = np.ones((10, 10, 2), np.float32)
+    out = resize_flow(flow, 20, 5)
+    assert out.shape == (5, 20, 2)
+    assert abs(out[2, 10, 0] - 2.0) < 1e-5   # x doubled
+    assert abs(out[2, 10, 1] - 0.5) < 1e-5   # y halved
+
+
+def test_colorwheel():
+    w = make_colorwheel()
+    assert w.shape == (55, 3)
+    assert w[0].tolist() == [255.0, 0.0, 0.0]        # pure red start
+    assert w.max() == 255 and w.min() == 0
+
+
+def test_flow_to_color_zero_flow_is_white():
+    img = flow_to_color(np.zeros((4, 4, 2), np.float32))
+    assert img.shape == (4, 4, 3)
+    assert (img == 255).all()
+
+
+def test_flow_to_color_bgr_flag():
+    flow = np.zeros((4, 4, 2), np.float32)
+    flow[:, :, 0] = 1.0
+    rgb = flow_to_color(flow, convert_to_bgr=False)
+    bgr = flow_to_color(flow, convert_to_bgr=True)
+    assert np.array_equal(rgb[..., ::-1], bgr)
+
+
+def test_synthetic_pair_ground_truth():
+    im1, im2, flow = synthetic_pair(1, 64, 96, seed=3)
+    assert im1.shape == (1, 3, 64, 96) and flow.shape == (1, 2, 64, 96)
+    assert not torch.allclose(im1, im2)
+    # warping im2 back by the flow must approximately recover im1
+    rec = warp(im2, flow)
+    err_moved = (rec - im1).abs().mean()
+    err_raw = (im2 - im1).abs().mean()
+    assert err_moved < 0.5 * err_raw
+
+
+def test_pad8_unpad():
+    x = torch.rand(1, 3, 61, 99)
+    p, hw = pad8(x)
+    assert p.shape[-2:] == (64, 104)
+    assert hw == (61, 99)
+    assert torch.equal(unpad(p, hw), x)
+
+
+def test_inference_engine_dynamic_shapes():
+    from raft_amd import RAFT, RaftConfig
+    eng = InferenceEngine(RAFT(RaftConfig(small=True)), iters=2)
+    for h, w in [(61, 99), (64, 96)]:
+        out = eng(torch.rand(1, 3, h, w), torch.rand(1, 3, h, w))
+        assert out.shape == (1, 2, h, w)
+
+
+def test_dataflow_resize_and_batch(tmp_path):
+    img = (np.random.rand(40, 60, 3) * 255).astype(np.uint8)
+    paths = []
+    for i in range(3):
+        p = str(tmp_path / f"f{i}.png")
+        write_png(p, img)
+        paths.append(p)
+    from raft_amd.data.dataflow import PairDataflow
+    ds = PairDataflow([(paths[0], paths[1]), (paths[1], paths[2])],
+                      input_size=(32, 48), batch=2)
+    batches = list(ds)
+    assert len(batches) == 1
+    b1, b2 = batches[0]
+    assert b1.shape == (2, 3, 32, 48)
+    assert b1.dtype == torch.float32 and b1.max() <= 1.0
+
+
+def test_cli_flops_mode(capsys):
+    import infer_raft
+    infer_raft.main(["--mode", "flops", "--small", "--iters", "4"])
+    out = capsys.readouterr().out
+    assert '"params": 990162' in out
+
+
+def test_cli_export_mode(tmp_path):
+    import infer_raft
+    infer_raft.main(["--mode", "export", "--small", "--out", str(tmp_path)])
+    assert os.path.exists(tmp_path / "raft-small.npz")
+    # exported npz reloads
+    infer_raft.main(["--mode", "flops", "--small"])
