@@ -147,6 +147,15 @@ def main():
 
 def _make_infer_step(model, x1, x2, args):
     iters = args.iters
+    if not args.no_graph and torch.cuda.is_available():
+        from raft_amd.engine.inference import InferenceEngine
+        engine = InferenceEngine(model, iters=iters, dtype=x1.dtype,
+                                 use_graph=True)
+
+        def step():
+            return engine(x1, x2)
+
+        return step, lambda: None
 
     def step():
         with torch.no_grad():

@@ -1,0 +1,245 @@
+"""Fused NHWC bf16 inference path (MI355X-native hot loop).
+
+Runs the whole RAFT refinement loop on the hand-written gfx950 kernels
+(raft_amd/ops/csrc/{corr_nhwc,fconv}.hip), operating on *physical NHWC*
+tensors throughout:
+
+  encoders (PyTorch/MIOpen, channels-last)  ->  bf16 NT-GEMM corr volume
+  -> bf16 pooled pyramid -> per iteration: NHWC lookup (channel-padded) ->
+  fused motion encoder (5 fconvs writing into the reusable GRU input
+  buffer) -> fused SepConvGRU (2 fconv pairs per direction with the gate
+  math in the epilogue) -> fused flow head + mask head -> coords update
+  -> final convex upsample.
+
+This path is inference-only (no autograd) and numerics-matched to the
+eager model (tests/test_fused.py compares against the fp32 golden path).
+Training keeps the autograd ops.
+
+Weight packing: conv weights [N, Cin, kh, kw] -> [kh*kw, N, Cin_pad] bf16
+(c-contiguous = the MFMA B-fragment order); the mask head's x0.25 scale
+(model_utils.py:183) is folded into its packed weights and bias.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+ACT_NONE, ACT_RELU, ACT_SIGMOID, ACT_TANH = 0, 1, 2, 3
+
+
+def _pad8(c: int) -> int:
+    return (c + 7) // 8 * 8
+
+
+def pack_conv(conv: torch.nn.Conv2d, pad_cin: Optional[int] = None,
+              scale: float = 1.0):
+    """[N, Cin, kh, kw] -> ([taps, N, Cin_pad] bf16 contiguous, bias fp32)."""
+    w = conv.weight.detach().float()
+    N, Cin, kh, kw = w.shape
+    cp = pad_cin if pad_cin is not None else Cin
+    wp = w.new_zeros(N, cp, kh, kw)
+    wp[:, :Cin] = w * scale
+    wp = wp.permute(2, 3, 0, 1).reshape(kh * kw, N, cp)
+    wp = wp.contiguous().to(device=conv.weight.device, dtype=torch.bfloat16)
+    if conv.bias is not None:
+        bias = (conv.bias.detach().float() * scale).contiguous()
+    else:
+        bias = torch.zeros(N, device=conv.weight.device)
+    return wp, bias, kh, kw
+
+
+def pack_zr(convz: torch.nn.Conv2d, convr: torch.nn.Conv2d):
+    """Stack [Wz; Wr] along N for the fused z/r gate kernel."""
+    wz, bz, kh, kw = pack_conv(convz)
+    wr, br, _, _ = pack_conv(convr)
+    return torch.cat([wz, wr], dim=1).contiguous(), \
+        torch.cat([bz, br]).contiguous(), kh, kw
+
+
+class _FC:
+    """A packed conv ready for fconv_plain."""
+
+    def __init__(self, conv, pad_cin=None, scale=1.0):
+        self.wp, self.bias, self.kh, self.kw = pack_conv(conv, pad_cin, scale)
+
+    def __call__(self, hip, in1, in2=None, act=ACT_RELU, out=None, n_off=0):
+        return hip.fconv_plain(in1, in2, self.wp, self.bias, self.kh,
+                               self.kw, act, out, n_off)
+
+
+class _GruDir:
+    def __init__(self, convz, convr, convq):
+        self.zr_w, self.zr_b, self.kh, self.kw = pack_zr(convz, convr)
+        self.q_w, self.q_b, _, _ = pack_conv(convq)
+
+    def __call__(self, hip, h, x):
+        z, rh = hip.fconv_gru_zr(h, x, self.zr_w, self.zr_b, self.kh, self.kw)
+        return hip.fconv_gru_q(rh, x, self.q_w, self.q_b, self.kh, self.kw,
+                               z, h)
+
+
+class FusedBasicUpdate:
+    """Packed raft-things update block (motion enc + SepConvGRU + heads)."""
+
+    def __init__(self, ub, corr_cpad: int, ctx_dim: int):
+        enc = ub.encoder
+        self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
+        self.c2 = _FC(enc.convc2)
+        self.f1 = _FC(enc.convf1)
+        self.f2 = _FC(enc.convf2)
+        self.cv = _FC(enc.conv)                      # in: [cor(192)|flo(64)]
+        self.gru1 = _GruDir(ub.gru.convz1, ub.gru.convr1, ub.gru.convq1)
+        self.gru2 = _GruDir(ub.gru.convz2, ub.gru.convr2, ub.gru.convq2)
+        self.fh1 = _FC(ub.flow_head.conv1)
+        self.fh2 = _FC(ub.flow_head.conv2)
+        self.m0 = _FC(ub.mask[0])
+        self.m2 = _FC(ub.mask[2], scale=0.25)        # fold the 0.25 scale
+        self.ctx_dim = ctx_dim
+
+    def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
+        # motion encoder (model_utils.py:110-119), outputs into x_buf slice
+        cor = self.c2(hip, self.c1(hip, corr_pad))
+        flo = self.f2(hip, self.f1(hip, flow_bf))
+        ctx = self.ctx_dim
+        self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
+        x_buf[..., ctx + 126:ctx + 128] = flow_bf
+        # SepConvGRU (model_utils.py:138-156)
+        net = self.gru1(hip, net, x_buf)
+        net = self.gru2(hip, net, x_buf)
+        # heads
+        dflow = self.fh2(hip, self.fh1(hip, net), act=ACT_NONE)
+        mask = self.m2(hip, self.m0(hip, net), act=ACT_NONE)
+        return net, mask, dflow
+
+
+class FusedSmallUpdate:
+    """Packed raft-small update block (motion enc + ConvGRU, no mask)."""
+
+    def __init__(self, ub, corr_cpad: int, ctx_dim: int):
+        enc = ub.encoder
+        self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
+        self.f1 = _FC(enc.convf1)
+        self.f2 = _FC(enc.convf2)
+        self.cv = _FC(enc.conv)                      # in: [cor(96)|flo(32)]
+        self.gru = _GruDir(ub.gru.convz, ub.gru.convr, ub.gru.convq)
+        self.fh1 = _FC(ub.flow_head.conv1)
+        self.fh2 = _FC(ub.flow_head.conv2)
+        self.ctx_dim = ctx_dim
+        # x = [inp(ctx) | motion(80) | flow(2)]; motion encoder out = 80
+
+    def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
+        cor = self.c1(hip, corr_pad)
+        flo = self.f2(hip, self.f1(hip, flow_bf))
+        ctx = self.ctx_dim
+        self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
+        x_buf[..., ctx + 80:ctx + 82] = flow_bf
+        net = self.gru(hip, net, x_buf)
+        dflow = self.fh2(hip, self.fh1(hip, net), act=ACT_NONE)
+        return net, None, dflow
+
+
+class FusedRaft:
+    """Caches packed weights for a RAFT model; call run() for inference."""
+
+    def __init__(self, model):
+        from raft_amd.ops import require_hip
+        self.hip = require_hip()
+        self.model = model
+        cfg = model.cfg
+        K = 2 * cfg.corr_radius + 1
+        self.corr_c = cfg.corr_levels * K * K
+        self.corr_cpad = _pad8(self.corr_c)
+        if cfg.small:
+            self.update = FusedSmallUpdate(model.update_block, self.corr_cpad,
+                                           cfg.context_dim)
+            self.x_dim = cfg.context_dim + 82
+        else:
+            self.update = FusedBasicUpdate(model.update_block, self.corr_cpad,
+                                           cfg.context_dim)
+            self.x_dim = cfg.context_dim + 128
+        self._version = self._weights_version()
+
+    def _weights_version(self) -> int:
+        return sum(p._version for p in self.model.update_block.parameters())
+
+    def stale(self) -> bool:
+        return self._version != self._weights_version()
+
+    @torch.no_grad()
+    def run(self, image1, image2, iters, flow_init=None):
+        model = self.model
+        cfg = model.cfg
+        hip = self.hip
+        img1 = model.preprocess(image1)
+        img2 = model.preprocess(image2)
+        fmaps = model.fnet(torch.cat([img1, img2], dim=0))
+        fmap1, fmap2 = torch.chunk(fmaps, 2, dim=0)
+        # physical NHWC views ([B,C,H,W] channels-last -> permute is free)
+        f1p = fmap1.permute(0, 2, 3, 1).contiguous()
+        f2p = fmap2.permute(0, 2, 3, 1).contiguous()
+        B, H8, W8, C = f1p.shape
+
+        vol = hip.corr_volume_nhwc(f1p, f2p, True)       # bf16 volume
+        levels = [vol]
+        for _ in range(cfg.corr_levels - 1):
+            levels.append(hip.corr_pool2x_bf16(levels[-1]))
+
+        cnet = model.cnet(img1)
+        net, inp = torch.split(cnet, [cfg.hidden_dim, cfg.context_dim], dim=1)
+        net = torch.tanh(net).permute(0, 2, 3, 1).contiguous()
+        inp = torch.relu(inp).permute(0, 2, 3, 1).contiguous()
+
+        x_buf = torch.empty(B, H8, W8, self.x_dim, device=net.device,
+                            dtype=torch.bfloat16)
+        x_buf[..., :cfg.context_dim] = inp
+
+        ys, xs = torch.meshgrid(
+            torch.arange(H8, device=net.device, dtype=torch.float32),
+            torch.arange(W8, device=net.device, dtype=torch.float32),
+            indexing="ij")
+        coords0 = torch.stack([xs, ys], dim=-1)[None].expand(B, -1, -1, -1) \
+            .contiguous()                                # [B,H,W,2] (x,y)
+        coords1 = coords0.clone()
+        if flow_init is not None:                        # [B,2,H,W] logical
+            coords1 = coords1 + flow_init.permute(0, 2, 3, 1).float()
+
+        mask = None
+        for _ in range(iters):
+            corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
+                                            cfg.corr_radius, self.corr_cpad,
+                                            True)
+            flow_bf = (coords1 - coords0).to(torch.bfloat16)
+            net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
+            coords1 = coords1 + dflow.float()
+
+        flow = (coords1 - coords0)                       # [B,H,W,2] fp32
+        if cfg.small:
+            from raft_amd.ops import torch_ref
+            up = torch_ref.upflow8(flow.permute(0, 3, 1, 2))
+            if cfg.scale_small_upflow:
+                up = 8.0 * up
+            return up
+        flow_nchw = flow.permute(0, 3, 1, 2).contiguous()
+        mask_nchw = mask.permute(0, 3, 1, 2).contiguous()
+        return hip.convex_upsample(flow_nchw, mask_nchw)
+
+
+def get_fused(model) -> Optional[FusedRaft]:
+    """Return (building if needed) the packed fused runner for this model.
+    Rebuilds automatically when the underlying weights changed."""
+    f = getattr(model, "_fused_cache", None)
+    if f is None or f.stale():
+        f = FusedRaft(model)
+        model._fused_cache = f
+    return f
+
+
+def can_fuse(model, image1: torch.Tensor) -> bool:
+    import raft_amd.ops as O
+    try:
+        p = next(model.update_block.parameters())
+    except StopIteration:
+        return False
+    return (image1.is_cuda and not torch.is_grad_enabled()
+            and p.dtype == torch.bfloat16 and O.hip_available())

@@ -107,6 +107,15 @@ class RAFT(nn.Module):
         training path).
         """
         iters = iters if iters is not None else self.cfg.iters
+
+        # MI355X fast path: whole refinement loop on the fused NHWC bf16
+        # kernels (inference only; numerics-tested vs this eager path).
+        if test_mode:
+            from raft_amd.models import fused
+            if fused.can_fuse(self, image1):
+                return fused.get_fused(self).run(image1, image2, iters,
+                                                 flow_init)
+
         img1 = self.preprocess(image1)
         img2 = self.preprocess(image2)
 

@@ -32,6 +32,20 @@ void launch_convex_upsample_fwd_f32(const float*, const float*, float*, int,
 void launch_convex_upsample_bwd_f32(const float*, const float*, const float*,
                                     float*, float*, int, int, int,
                                     hipStream_t);
+void launch_corr_volume_nhwc_bf16(const void*, const void*, void*, bool,
+                                  int, int, int, int, float, hipStream_t);
+void launch_corr_lookup_nhwc(const void* const*, const int*, const int*,
+                             bool, const float*, void*, bool, int, int, int,
+                             int, int, int, hipStream_t);
+void launch_corr_lookup_nhwc_bwd(float* const*, const int*, const int*,
+                                 const float*, const float*, int, int, int,
+                                 int, int, hipStream_t);
+void launch_corr_pool2x_bf16(const void*, void*, int, int, int, int,
+                             long long, hipStream_t);
+void launch_fconv_nhwc_bf16(const void*, int, const void*, int, const void*,
+                            const float*, void*, int, int, int, int, int,
+                            int, int, int, int, int, const void*,
+                            const void*, void*, void*, hipStream_t);
 }
 
 namespace {
@@ -181,6 +195,148 @@ std::vector<at::Tensor> convex_upsample_backward(at::Tensor grad_up,
     return {grad_flow.to(flow.scalar_type()), grad_mask.to(mask.scalar_type())};
 }
 
+// --------------------------------------------------------------- NHWC path
+// Physical-NHWC tensors ([B,H,W,C] row-major) for the fused inference loop.
+
+at::Tensor corr_volume_nhwc(at::Tensor f1, at::Tensor f2, bool out_bf16) {
+    // f1/f2: physical NHWC [B, H, W, C] bf16, row-major contiguous
+    CHECK_DEV(f1); CHECK_CONT(f1); CHECK_DEV(f2); CHECK_CONT(f2);
+    TORCH_CHECK(f1.scalar_type() == at::kBFloat16, "nhwc volume needs bf16");
+    TORCH_CHECK(f1.sizes() == f2.sizes());
+    const int B = f1.size(0), H = f1.size(1), W = f1.size(2), C = f1.size(3);
+    TORCH_CHECK(C % 64 == 0, "C must be a multiple of 64");
+    const int M = H * W;
+    auto out = at::empty({B, M, H, W},
+                         f1.options().dtype(out_bf16 ? at::kBFloat16
+                                                     : at::kFloat));
+    launch_corr_volume_nhwc_bf16(f1.data_ptr(), f2.data_ptr(),
+                                 out.data_ptr(), out_bf16, B, M, M, C,
+                                 1.0f / std::sqrt((float)C),
+                                 current_stream());
+    return out;
+}
+
+at::Tensor corr_pool2x_bf16(at::Tensor corr) {
+    CHECK_DEV(corr); CHECK_CONT(corr);
+    TORCH_CHECK(corr.scalar_type() == at::kBFloat16);
+    const int B = corr.size(0), Q = corr.size(1);
+    const int H = corr.size(2), W = corr.size(3);
+    const int Ho = H / 2, Wo = W / 2;
+    auto out = at::empty({B, Q, Ho, Wo}, corr.options());
+    launch_corr_pool2x_bf16(corr.data_ptr(), out.data_ptr(), H, W, Ho, Wo,
+                            (long long)B * Q * Ho * Wo, current_stream());
+    return out;
+}
+
+at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
+                            at::Tensor coords, int64_t radius,
+                            int64_t c_stride, bool out_bf16) {
+    // coords: [B, H, W, 2] fp32; returns physical NHWC [B, H, W, c_stride]
+    // with channels [L*KK..c_stride) zero-filled (pad for fconv Cin%8).
+    CHECK_DEV(coords); CHECK_CONT(coords);
+    const int B = coords.size(0), H = coords.size(1), W = coords.size(2);
+    const int L = (int)levels.size();
+    const int K = 2 * (int)radius + 1;
+    const int C = L * K * K;
+    TORCH_CHECK(c_stride >= C);
+    const void* ptrs[4];
+    int hs[4] = {0}, ws[4] = {0};
+    bool vol_bf16 = levels[0].scalar_type() == at::kBFloat16;
+    for (int i = 0; i < L; ++i) {
+        CHECK_DEV(levels[i]); CHECK_CONT(levels[i]);
+        ptrs[i] = levels[i].data_ptr();
+        hs[i] = levels[i].size(2);
+        ws[i] = levels[i].size(3);
+    }
+    // pad channels [C..c_stride) must be zero for the consumer's padded
+    // weights; zero-fill once per call only when padded (kernel writes the
+    // first C channels of every row with the Cs stride).
+    auto opts = coords.options().dtype(out_bf16 ? at::kBFloat16
+                                                 : at::kFloat);
+    auto out = (c_stride == C) ? at::empty({B, H, W, (int64_t)c_stride}, opts)
+                               : at::zeros({B, H, W, (int64_t)c_stride}, opts);
+    launch_corr_lookup_nhwc(ptrs, hs, ws, vol_bf16,
+                            coords.data_ptr<float>(), out.data_ptr(),
+                            out_bf16, B, H, W, L, (int)radius,
+                            (int)c_stride, current_stream());
+    return out;
+}
+
+at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
+                       at::Tensor wp, c10::optional<at::Tensor> bias,
+                       int64_t kh, int64_t kw, int64_t act,
+                       c10::optional<at::Tensor> out_buf, int64_t n_off) {
+    CHECK_DEV(in1); CHECK_CONT(in1); CHECK_DEV(wp); CHECK_CONT(wp);
+    TORCH_CHECK(in1.scalar_type() == at::kBFloat16, "fconv needs bf16");
+    const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
+    const int C1 = in1.size(3);
+    const void* p2 = nullptr;
+    int C2 = 0;
+    if (in2.has_value()) {
+        CHECK_CONT(in2.value());
+        p2 = in2->data_ptr();
+        C2 = in2->size(3);
+    }
+    const int N = wp.size(1);
+    TORCH_CHECK(wp.size(0) == kh * kw && wp.size(2) == C1 + C2,
+                "packed weight shape mismatch");
+    const float* bptr = nullptr;
+    if (bias.has_value()) {
+        TORCH_CHECK(bias->scalar_type() == at::kFloat);
+        bptr = bias->data_ptr<float>();
+    }
+    at::Tensor out;
+    int cstride;
+    if (out_buf.has_value()) {
+        out = out_buf.value();
+        CHECK_CONT(out);
+        cstride = out.size(3);
+    } else {
+        out = at::empty({B, H, W, N}, in1.options());
+        cstride = N;
+        n_off = 0;
+    }
+    launch_fconv_nhwc_bf16(in1.data_ptr(), C1, p2, C2, wp.data_ptr(), bptr,
+                           out.data_ptr(), B, H, W, N, (int)n_off, cstride,
+                           (int)kh, (int)kw, (int)act, 0, nullptr, nullptr,
+                           nullptr, nullptr, current_stream());
+    return out;
+}
+
+std::vector<at::Tensor> fconv_gru_zr(at::Tensor h, at::Tensor x,
+                                     at::Tensor wp, at::Tensor bias,
+                                     int64_t kh, int64_t kw) {
+    CHECK_DEV(h); CHECK_CONT(h); CHECK_CONT(x); CHECK_CONT(wp);
+    const int B = h.size(0), H = h.size(1), W = h.size(2), hd = h.size(3);
+    const int N = wp.size(1);
+    TORCH_CHECK(N == 2 * hd, "zr weights must stack [Wz; Wr]");
+    auto z = at::empty_like(h);
+    auto rh = at::empty_like(h);
+    launch_fconv_nhwc_bf16(h.data_ptr(), hd, x.data_ptr(), x.size(3),
+                           wp.data_ptr(), bias.data_ptr<float>(), nullptr,
+                           B, H, W, N, 0, 0, (int)kh, (int)kw, 0, 1,
+                           h.data_ptr(), nullptr, z.data_ptr(),
+                           rh.data_ptr(), current_stream());
+    return {z, rh};
+}
+
+at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
+                       at::Tensor bias, int64_t kh, int64_t kw, at::Tensor z,
+                       at::Tensor h) {
+    CHECK_DEV(rh); CHECK_CONT(rh); CHECK_CONT(x); CHECK_CONT(wp);
+    CHECK_CONT(z); CHECK_CONT(h);
+    const int B = rh.size(0), H = rh.size(1), W = rh.size(2);
+    const int hd = rh.size(3);
+    TORCH_CHECK(wp.size(1) == hd);
+    auto out = at::empty_like(h);
+    launch_fconv_nhwc_bf16(rh.data_ptr(), hd, x.data_ptr(), x.size(3),
+                           wp.data_ptr(), bias.data_ptr<float>(),
+                           out.data_ptr(), B, H, W, hd, 0, hd, (int)kh,
+                           (int)kw, 0, 2, h.data_ptr(), z.data_ptr(),
+                           nullptr, nullptr, current_stream());
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -192,4 +348,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_gates_bwd", &gru_gates_bwd);
     m.def("convex_upsample", &convex_upsample, "8x convex upsample");
     m.def("convex_upsample_backward", &convex_upsample_backward);
+    // NHWC / fused inference path
+    m.def("corr_volume_nhwc", &corr_volume_nhwc,
+          "bf16 NT-GEMM correlation volume (physical NHWC fmaps)");
+    m.def("corr_pool2x_bf16", &corr_pool2x_bf16);
+    m.def("corr_lookup_nhwc", &corr_lookup_nhwc,
+          "pyramid lookup writing physical NHWC (channel-padded)");
+    m.def("fconv_plain", &fconv_plain,
+          "fused NHWC bf16 conv (+bias +activation, slice output)");
+    m.def("fconv_gru_zr", &fconv_gru_zr, "GRU z/r gate conv pair");
+    m.def("fconv_gru_q", &fconv_gru_q, "GRU candidate conv + state update");
 }

@@ -1,0 +1,279 @@
+// NHWC (channels-last) correlation kernels — the MI355X-native layout.
+//
+// corr_volume_nhwc: C[b,m,n] = sum_k F1[b,m,k] * F2[b,n,k] / sqrt(K), bf16
+// inputs, fp32 accumulate (MFMA v_mfma_f32_16x16x32_bf16), fp32 or bf16
+// volume out.  Channels-last fmaps make BOTH operands k-contiguous, so LDS
+// staging is a straight contiguous copy and fragments are ds_read_b128 with
+// an XOR swizzle (guide §5.5 T2) — no transposes anywhere.
+//
+// corr_lookup_nhwc: same tap math as corr_lookup.hip (edge-clamp trunc
+// bilinear, [::-1] window order) but the output is physical NHWC
+// [B,H,W,L*KK] with the tap channel fastest: consecutive lanes write
+// consecutive channels of one query (fully coalesced) and read overlapping
+// window cells (L1-friendly).  Template on the volume dtype (fp32 / bf16).
+
+#include "common.h"
+#include <type_traits>
+#include <hip/hip_bf16.h>
+
+typedef short short8 __attribute__((ext_vector_type(8)));
+typedef unsigned int uint4v __attribute__((ext_vector_type(4)));
+
+#define CV_BM 128
+#define CV_BN 128
+#define CV_BK 64
+// +16 B row pad: consecutive rows land on distinct 16-B bank slots mod the
+// 256-B bank row (stride 144: 16 distinct residues), so the 16-lane
+// ds_read_b128 groups (16 different rows, same column) are conflict-free
+// (guide §2/§6 Guideline 4).
+#define CV_ROWB (CV_BK * 2 + 16)
+
+RAFT_DEV unsigned swz(int row, unsigned colbyte) {
+    return row * CV_ROWB + colbyte;
+}
+
+template <typename OUT_T>
+__global__ __launch_bounds__(256) void corr_volume_nhwc_bf16_k(
+    const __hip_bfloat16* __restrict__ f1,   // [B, M, K]
+    const __hip_bfloat16* __restrict__ f2,   // [B, N, K]
+    OUT_T* __restrict__ out,                 // [B, M, N]
+    int M, int N, int K, float scale) {
+    __shared__ char smem[2 * CV_BM * CV_ROWB];
+    char* sA = smem;
+    char* sB = smem + CV_BM * CV_ROWB;
+
+    const int b = blockIdx.z;
+    const int m0 = blockIdx.y * CV_BM;
+    const int n0 = blockIdx.x * CV_BN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = (wave >> 1) * 64;
+    const int wn = (wave & 1) * 64;
+
+    const __hip_bfloat16* A = f1 + (size_t)b * M * K;
+    const __hip_bfloat16* Bp = f2 + (size_t)b * N * K;
+
+    floatx4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int srow = tid >> 1;                 // 0..127
+    const unsigned scol = (tid & 1) * 64;      // byte offset within row half
+
+    for (int k0 = 0; k0 < K; k0 += CV_BK) {
+        // stage: each thread copies 64 B (32 bf16) of one row, 4x16B chunks
+        {
+            const int ma = m0 + srow;
+            const int na = n0 + srow;
+            const __hip_bfloat16* ga =
+                A + (size_t)min(ma, M - 1) * K + k0 + (tid & 1) * 32;
+            const __hip_bfloat16* gb =
+                Bp + (size_t)min(na, N - 1) * K + k0 + (tid & 1) * 32;
+            const bool oka = ma < M, okb = na < N;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+                uint4v va = oka ? *(const uint4v*)(ga + j * 8)
+                                : uint4v{0, 0, 0, 0};
+                uint4v vb = okb ? *(const uint4v*)(gb + j * 8)
+                                : uint4v{0, 0, 0, 0};
+                *(uint4v*)(sA + swz(srow, scol + j * 16)) = va;
+                *(uint4v*)(sB + swz(srow, scol + j * 16)) = vb;
+            }
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < CV_BK / 32; ++kk) {
+            short8 af[4], bf[4];
+            const unsigned cb = kk * 64 + (lane >> 4) * 16;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                af[i] = *(const short8*)(sA + swz(wm + i * 16 + (lane & 15), cb));
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+                bf[j] = *(const short8*)(sB + swz(wn + j * 16 + (lane & 15), cb));
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = m0 + wm + i * 16 + (lane >> 4) * 4 + r;
+                const int n = n0 + wn + j * 16 + (lane & 15);
+                if (m < M && n < N)
+                    out[((size_t)b * M + m) * N + n] =
+                        (OUT_T)(acc[i][j][r] * scale);
+            }
+}
+
+extern "C" void launch_corr_volume_nhwc_bf16(
+    const void* f1, const void* f2, void* out, bool out_bf16, int Bsz,
+    int M, int N, int K, float scale, hipStream_t s) {
+    dim3 grid(cdiv(N, CV_BN), cdiv(M, CV_BM), Bsz);
+    if (out_bf16)
+        hipLaunchKernelGGL(corr_volume_nhwc_bf16_k<__hip_bfloat16>, grid,
+                           dim3(256), 0, s, (const __hip_bfloat16*)f1,
+                           (const __hip_bfloat16*)f2, (__hip_bfloat16*)out,
+                           M, N, K, scale);
+    else
+        hipLaunchKernelGGL(corr_volume_nhwc_bf16_k<float>, grid, dim3(256),
+                           0, s, (const __hip_bfloat16*)f1,
+                           (const __hip_bfloat16*)f2, (float*)out,
+                           M, N, K, scale);
+}
+
+// --------------------------------------------------------------- NHWC lookup
+struct LevelsT {
+    const void* ptr[4];
+    int H[4];
+    int W[4];
+};
+
+template <typename T, typename OT>
+__global__ void corr_lookup_nhwc_k(
+    LevelsT lv, const float* __restrict__ coords,  // [B, H, W, 2]
+    OT* __restrict__ out,                          // [B, H, W, Cs]
+    int H, int W, int num_levels, int radius, int Cs, long long total) {
+    const int K = 2 * radius + 1;
+    const int KK = K * K;
+    const int C = num_levels * KK;
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const int c = (int)(idx % C);
+        const long long q = idx / C;               // b*H*W + y*W + x
+        const int lvl = c / KK;
+        const int k = c - lvl * KK;
+
+        const float inv = 1.0f / (float)(1 << lvl);
+        const float cx = coords[q * 2] * inv + (float)(k / K - radius);
+        const float cy = coords[q * 2 + 1] * inv + (float)(k % K - radius);
+
+        const int H2 = lv.H[lvl], W2 = lv.W[lvl];
+        const T* slice = (const T*)lv.ptr[lvl] + q * (size_t)H2 * W2;
+        BilinearTap t = make_tap(cx, cy, W2, H2);
+        const float Ia = (float)slice[t.y0 * W2 + t.x0];
+        const float Ib = (float)slice[t.y1 * W2 + t.x0];
+        const float Ic = (float)slice[t.y0 * W2 + t.x1];
+        const float Id = (float)slice[t.y1 * W2 + t.x1];
+        out[q * Cs + c] = (OT)(t.wa * Ia + t.wb * Ib + t.wc * Ic + t.wd * Id);
+    }
+}
+
+extern "C" void launch_corr_lookup_nhwc(
+    const void* const* level_ptrs, const int* level_h, const int* level_w,
+    bool vol_bf16, const float* coords, void* out, bool out_bf16, int B,
+    int H, int W, int num_levels, int radius, int Cs, hipStream_t s) {
+    LevelsT lv{};
+    for (int i = 0; i < num_levels; ++i) {
+        lv.ptr[i] = level_ptrs[i];
+        lv.H[i] = level_h[i];
+        lv.W[i] = level_w[i];
+    }
+    const int K = 2 * radius + 1;
+    const long long total = (long long)B * H * W * num_levels * K * K;
+    int blocks = (int)min((total + 255) / 256, (long long)8192);
+#define LKCASE(T, OT, OTC)                                                  \
+    if (vol_bf16 == std::is_same<T, __hip_bfloat16>::value &&               \
+        out_bf16 == OTC) {                                                  \
+        hipLaunchKernelGGL((corr_lookup_nhwc_k<T, OT>), dim3(blocks),       \
+                           dim3(256), 0, s, lv, coords, (OT*)out, H, W,     \
+                           num_levels, radius, Cs, total);                  \
+        return;                                                             \
+    }
+    LKCASE(float, float, false)
+    LKCASE(float, __hip_bfloat16, true)
+    LKCASE(__hip_bfloat16, float, false)
+    LKCASE(__hip_bfloat16, __hip_bfloat16, true)
+#undef LKCASE
+}
+
+// backward: scatter into fp32 grad volumes from an NHWC grad_out
+struct GradLevels {
+    float* ptr[4];
+    int H[4];
+    int W[4];
+};
+
+extern "C" __global__ void corr_lookup_nhwc_bwd_k(
+    GradLevels lv, const float* __restrict__ coords,
+    const float* __restrict__ grad_out,            // [B, H, W, L*KK]
+    int H, int W, int num_levels, int radius, long long total) {
+    const int K = 2 * radius + 1;
+    const int KK = K * K;
+    const int C = num_levels * KK;
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const float g = grad_out[idx];
+        const int c = (int)(idx % C);
+        const long long q = idx / C;
+        const int lvl = c / KK;
+        const int k = c - lvl * KK;
+        const float inv = 1.0f / (float)(1 << lvl);
+        const float cx = coords[q * 2] * inv + (float)(k / K - radius);
+        const float cy = coords[q * 2 + 1] * inv + (float)(k % K - radius);
+        const int H2 = lv.H[lvl], W2 = lv.W[lvl];
+        float* slice = lv.ptr[lvl] + q * (size_t)H2 * W2;
+        BilinearTap t = make_tap(cx, cy, W2, H2);
+        atomicAdd(&slice[t.y0 * W2 + t.x0], t.wa * g);
+        atomicAdd(&slice[t.y1 * W2 + t.x0], t.wb * g);
+        atomicAdd(&slice[t.y0 * W2 + t.x1], t.wc * g);
+        atomicAdd(&slice[t.y1 * W2 + t.x1], t.wd * g);
+    }
+}
+
+extern "C" void launch_corr_lookup_nhwc_bwd(
+    float* const* grad_ptrs, const int* level_h, const int* level_w,
+    const float* coords, const float* grad_out, int B, int H, int W,
+    int num_levels, int radius, hipStream_t s) {
+    GradLevels lv{};
+    for (int i = 0; i < num_levels; ++i) {
+        lv.ptr[i] = grad_ptrs[i];
+        lv.H[i] = level_h[i];
+        lv.W[i] = level_w[i];
+    }
+    const int K = 2 * radius + 1;
+    const long long total = (long long)B * H * W * num_levels * K * K;
+    int blocks = (int)min((total + 255) / 256, (long long)8192);
+    hipLaunchKernelGGL(corr_lookup_nhwc_bwd_k, dim3(blocks), dim3(256), 0, s,
+                       lv, coords, grad_out, H, W, num_levels, radius, total);
+}
+
+// bf16 2x2/2 avg pool (for the bf16 volume pyramid)
+extern "C" __global__ void corr_pool2x_bf16_k(
+    const __hip_bfloat16* __restrict__ in, __hip_bfloat16* __restrict__ out,
+    int H, int W, int Ho, int Wo, long long total) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const int xo = (int)(idx % Wo);
+        const int yo = (int)((idx / Wo) % Ho);
+        const long long q = idx / ((long long)Wo * Ho);
+        const __hip_bfloat16* src = in + (q * H + 2 * yo) * W + 2 * xo;
+        out[idx] = (__hip_bfloat16)(0.25f * ((float)src[0] + (float)src[1] +
+                                             (float)src[W] + (float)src[W + 1]));
+    }
+}
+
+extern "C" void launch_corr_pool2x_bf16(const void* in, void* out, int H,
+                                        int W, int Ho, int Wo,
+                                        long long total, hipStream_t s) {
+    int blocks = (int)min((total + 255) / 256, (long long)2048);
+    hipLaunchKernelGGL(corr_pool2x_bf16_k, dim3(blocks), dim3(256), 0, s,
+                       (const __hip_bfloat16*)in, (__hip_bfloat16*)out, H, W,
+                       Ho, Wo, total);
+}

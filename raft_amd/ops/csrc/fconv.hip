@@ -1,0 +1,252 @@
+// fconv: fused NHWC bf16 conv for the RAFT update loop (K4 and friends).
+//
+// Replaces the reference's per-iteration tensorpack Conv2D chains
+// (model_utils.py:110-185) — which PyTorch/MIOpen executed as
+// im2col + GEMM + transpose + bias + activation + concat kernels (~45% of
+// the step, see profiles/r01_eager_bench_kernel_stats.md) — with one
+// MFMA kernel per conv:
+//
+//   out[b,y,x,n] = act( sum_{ty,tx,c} In[b, y+ty-cy, x+tx-cx, c]
+//                                     * Wp[ty*kw+tx][n][c] + bias[n] )
+//
+//   * In is the logical channel-concat of up to TWO NHWC tensors (h|x for
+//     the GRU, cor|flo for the motion encoder) — no concat materialization;
+//   * Wp is host-packed [taps][Cout][Cin] (c contiguous = the MFMA B
+//     fragment order);
+//   * SAME zero padding, arbitrary odd kh x kw (templated per shape);
+//   * epilogue modes: plain activation / GRU z+r (sigmoid, r*h product) /
+//     GRU candidate (tanh + h' = (1-z)h + z*q) — model_utils.py:138-169;
+//   * output can be written into a channel-slice of a wider NHWC buffer
+//     (n_off / out_cstride), which is how the per-iteration GRU input
+//     buffer x = [ctx | motion | flow] is assembled without copies.
+//
+// Geometry: 256 threads = 4 waves (2m x 2n), tile BM=64 positions of one
+// output row x BN=128 channels; wave tile 32x64 = 2x4 fragments of
+// mfma_f32_16x16x32_bf16. K-loop: for each kernel row ty and BK=32 channel
+// step, stage the (BM + kw - 1)-wide input slab and all kw weight tiles,
+// one barrier pair, then kw shifted-LDS-read MFMA groups.
+
+#include "common.h"
+#include <hip/hip_bf16.h>
+
+typedef short short8 __attribute__((ext_vector_type(8)));
+typedef unsigned int uint4v __attribute__((ext_vector_type(4)));
+
+#define FC_BM 64
+#define FC_BN 128
+#define FC_BK 32
+// +16 B row pad -> 80-B stride: 16 consecutive rows hit 16 distinct 16-B
+// bank slots mod the 256-B bank row, so 16-lane ds_read_b128 groups (one
+// row per lane, same column) are conflict-free (guide §2/§6 Guideline 4).
+#define FC_ROWB (FC_BK * 2 + 16)
+
+RAFT_DEV unsigned fswz(int row, unsigned colbyte) {
+    return row * FC_ROWB + colbyte;
+}
+
+RAFT_DEV float factivate(float v, int act) {
+    if (act == 1) return fmaxf(v, 0.0f);
+    if (act == 2) return 1.0f / (1.0f + __expf(-v));
+    if (act == 3) return tanhf(v);
+    return v;
+}
+
+// epilogue modes
+#define EP_PLAIN 0
+#define EP_GRU_ZR 1   // N = 2*hd: [z | r] -> z_buf = sig(z), rh = sig(r)*h
+#define EP_GRU_Q 2    // N = hd: h' = (1-z)*h + z*tanh(q)
+
+template <int KH, int KW>
+__global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
+    const __hip_bfloat16* __restrict__ in1, int C1,
+    const __hip_bfloat16* __restrict__ in2, int C2,   // may be null/0
+    const __hip_bfloat16* __restrict__ wp,            // [KH*KW][N][C1+C2]
+    const float* __restrict__ bias,                   // [N] or null
+    __hip_bfloat16* __restrict__ out,                 // NHWC slice target
+    int H, int W, int N, int n_off, int out_cstride, int act, int mode,
+    const __hip_bfloat16* __restrict__ h_state,       // [B,H,W,hd] (GRU)
+    const __hip_bfloat16* __restrict__ z_buf_in,      // [B,H,W,hd] (EP_GRU_Q)
+    __hip_bfloat16* __restrict__ z_buf_out,           // [B,H,W,hd] (EP_GRU_ZR)
+    __hip_bfloat16* __restrict__ rh_out) {            // [B,H,W,hd] (EP_GRU_ZR)
+    constexpr int TAPS = KH * KW;
+    constexpr int AW = FC_BM + KW - 1;   // A slab width (positions + halo)
+    __shared__ char sA[AW * FC_ROWB];
+    __shared__ char sB[KW * FC_BN * FC_ROWB];
+
+    const int Cin = C1 + C2;
+    const int b = blockIdx.z;
+    const int tiles_per_row = (W + FC_BM - 1) / FC_BM;
+    const int y = blockIdx.y / tiles_per_row;
+    const int x0 = (blockIdx.y % tiles_per_row) * FC_BM;
+    const int n0 = blockIdx.x * FC_BN;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = (wave >> 1) * 32;
+    const int wn = (wave & 1) * 64;
+    const long long HW = (long long)H * W;
+
+    floatx4 acc[2][4];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+    for (int ty = 0; ty < KH; ++ty) {
+        const int row = y + ty - KH / 2;
+        const bool row_ok = (row >= 0 && row < H);
+        for (int k0 = 0; k0 < Cin; k0 += FC_BK) {
+            // ---- stage A slab: AW rows x FC_BK ch (8 bf16 per thread pass)
+            for (int e = tid; e < AW * (FC_BK / 8); e += 256) {
+                const int ar = e / (FC_BK / 8);
+                const int c8 = (e % (FC_BK / 8)) * 8;
+                const int x = x0 + ar - KW / 2;
+                uint4v v = {0, 0, 0, 0};
+                if (row_ok && x >= 0 && x < W) {
+                    const int k = k0 + c8;
+                    const long long p = ((long long)b * H + row) * W + x;
+                    if (k < C1) {
+                        if (k + 8 <= C1)
+                            v = *(const uint4v*)(in1 + p * C1 + k);
+                        else {  // straddles the in1|in2 seam: scalar gather
+                            __hip_bfloat16 tmp[8];
+                            for (int u = 0; u < 8; ++u) {
+                                const int kk = k + u;
+                                tmp[u] = (kk < C1) ? in1[p * C1 + kk]
+                                       : (kk - C1 < C2 ? in2[p * C2 + kk - C1]
+                                                       : (__hip_bfloat16)0.f);
+                            }
+                            v = *(const uint4v*)tmp;
+                        }
+                    } else if (k - C1 < C2) {
+                        if (k - C1 + 8 <= C2)
+                            v = *(const uint4v*)(in2 + p * C2 + (k - C1));
+                        else {
+                            __hip_bfloat16 tmp[8];
+                            for (int u = 0; u < 8; ++u) {
+                                const int kk = k - C1 + u;
+                                tmp[u] = kk < C2 ? in2[p * C2 + kk]
+                                                 : (__hip_bfloat16)0.f;
+                            }
+                            v = *(const uint4v*)tmp;
+                        }
+                    }
+                }
+                *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
+            }
+            // ---- stage all KW weight tiles [FC_BN][FC_BK]
+            for (int e = tid; e < KW * FC_BN * (FC_BK / 8); e += 256) {
+                const int t = e / (FC_BN * (FC_BK / 8));
+                const int rem = e % (FC_BN * (FC_BK / 8));
+                const int n = rem / (FC_BK / 8);
+                const int c8 = (rem % (FC_BK / 8)) * 8;
+                uint4v v = {0, 0, 0, 0};
+                const int gn = n0 + n;
+                const int k = k0 + c8;
+                if (gn < N && k + 8 <= Cin)
+                    v = *(const uint4v*)(
+                        wp + ((size_t)(ty * KW + t) * N + gn) * Cin + k);
+                else if (gn < N) {
+                    __hip_bfloat16 tmp[8];
+                    for (int u = 0; u < 8; ++u)
+                        tmp[u] = (k + u < Cin)
+                            ? wp[((size_t)(ty * KW + t) * N + gn) * Cin + k + u]
+                            : (__hip_bfloat16)0.f;
+                    v = *(const uint4v*)tmp;
+                }
+                *(uint4v*)(sB + t * FC_BN * FC_ROWB + fswz(n, c8 * 2)) = v;
+            }
+            __syncthreads();
+
+#pragma unroll
+            for (int tx = 0; tx < KW; ++tx) {
+                short8 af[2], bf[4];
+                const unsigned cb = (lane >> 4) * 16;
+#pragma unroll
+                for (int i = 0; i < 2; ++i)
+                    af[i] = *(const short8*)(
+                        sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+#pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    bf[j] = *(const short8*)(
+                        sB + tx * FC_BN * FC_ROWB +
+                        fswz(wn + j * 16 + (lane & 15), cb));
+#pragma unroll
+                for (int i = 0; i < 2; ++i)
+#pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af[i], bf[j], acc[i][j], 0, 0, 0);
+            }
+            __syncthreads();
+        }
+    }
+
+    // ------------------------------------------------------------- epilogue
+    const int hd = (mode == EP_GRU_ZR) ? N / 2 : N;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = wm + i * 16 + (lane >> 4) * 4 + r;
+                const int n = n0 + wn + j * 16 + (lane & 15);
+                const int x = x0 + m;
+                if (x >= W || n >= N) continue;
+                const long long p = ((long long)b * H + y) * W + x;
+                float v = acc[i][j][r];
+                if (bias) v += bias[n];
+                if (mode == EP_PLAIN) {
+                    out[p * out_cstride + n_off + n] =
+                        (__hip_bfloat16)factivate(v, act);
+                } else if (mode == EP_GRU_ZR) {
+                    const float s = 1.0f / (1.0f + __expf(-v));
+                    if (n < hd) {
+                        z_buf_out[p * hd + n] = (__hip_bfloat16)s;
+                    } else {
+                        const int c = n - hd;
+                        rh_out[p * hd + c] = (__hip_bfloat16)(
+                            s * (float)h_state[p * hd + c]);
+                    }
+                } else {  // EP_GRU_Q
+                    const float q = tanhf(v);
+                    const float z = (float)z_buf_in[p * hd + n];
+                    const float h = (float)h_state[p * hd + n];
+                    out[p * out_cstride + n_off + n] =
+                        (__hip_bfloat16)((1.0f - z) * h + z * q);
+                }
+            }
+}
+
+#define FCONV_ARGS                                                           \
+    (const __hip_bfloat16*)in1, C1, (const __hip_bfloat16*)in2, C2,          \
+    (const __hip_bfloat16*)wp, bias, (__hip_bfloat16*)out, H, W, N, n_off,   \
+    out_cstride, act, mode, (const __hip_bfloat16*)h_state,                  \
+    (const __hip_bfloat16*)z_buf_in, (__hip_bfloat16*)z_buf_out,             \
+    (__hip_bfloat16*)rh_out
+
+extern "C" void launch_fconv_nhwc_bf16(
+    const void* in1, int C1, const void* in2, int C2, const void* wp,
+    const float* bias, void* out, int B, int H, int W, int N, int n_off,
+    int out_cstride, int kh, int kw, int act, int mode, const void* h_state,
+    const void* z_buf_in, void* z_buf_out, void* rh_out, hipStream_t s) {
+    const int tiles_per_row = cdiv(W, FC_BM);
+    dim3 grid(cdiv(N, FC_BN), H * tiles_per_row, B);
+    dim3 blk(256);
+#define FC_CASE(KH, KW)                                                      \
+    if (kh == KH && kw == KW) {                                              \
+        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW>), grid, blk, 0, s,     \
+                           FCONV_ARGS);                                      \
+        return;                                                              \
+    }
+    FC_CASE(1, 1)
+    FC_CASE(3, 3)
+    FC_CASE(1, 5)
+    FC_CASE(5, 1)
+    FC_CASE(7, 7)
+    FC_CASE(1, 7)
+    FC_CASE(7, 1)
+#undef FC_CASE
+}

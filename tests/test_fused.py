@@ -1,0 +1,170 @@
+"""GPU numerics for the fused NHWC bf16 path (fconv, NHWC corr, full loop).
+
+References are the eager fp32/bf16 paths; tolerances account for bf16
+storage rounding (inputs and weights are bit-identical bf16 in both paths,
+and both accumulate in fp32, so differences are rounding-of-intermediates
+only).
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+from raft_amd.ops import torch_ref as R
+
+
+def _hip():
+    from raft_amd.ops import require_hip
+    return require_hip()
+
+
+@pytest.fixture
+def dev():
+    return torch.device("cuda:0")
+
+
+def _pack(w, scale=1.0, pad_cin=None):
+    N, Cin, kh, kw = w.shape
+    cp = pad_cin or Cin
+    wp = w.new_zeros(N, cp, kh, kw)
+    wp[:, :Cin] = w * scale
+    return wp.permute(2, 3, 0, 1).reshape(kh * kw, N, cp).contiguous() \
+        .to(torch.bfloat16)
+
+
+@pytest.mark.parametrize("kh,kw", [(1, 1), (3, 3), (1, 5), (5, 1), (7, 7)])
+def test_fconv_matches_conv2d(dev, kh, kw):
+    B, H, W, Cin, N = 2, 9, 21, 64, 96
+    x = torch.randn(B, H, W, Cin, device=dev).to(torch.bfloat16)
+    w = torch.randn(N, Cin, kh, kw, device=dev) * 0.1
+    bias = torch.randn(N, device=dev)
+    out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, kh, kw,
+                             1, None, 0)
+    # reference: fp32 conv on the bf16-rounded inputs/weights
+    xr = x.float().permute(0, 3, 1, 2)
+    ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias,
+                   padding=(kh // 2, kw // 2))
+    ref = F.relu(ref).permute(0, 2, 3, 1)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 0.05 * ref.abs().max().item() + 0.05, err
+
+
+def test_fconv_two_inputs_and_slice_output(dev):
+    B, H, W = 1, 6, 10
+    C1, C2, N = 32, 64, 48
+    a = torch.randn(B, H, W, C1, device=dev).to(torch.bfloat16)
+    b = torch.randn(B, H, W, C2, device=dev).to(torch.bfloat16)
+    w = torch.randn(N, C1 + C2, 3, 3, device=dev) * 0.1
+    bias = torch.zeros(N, device=dev)
+    buf = torch.zeros(B, H, W, 80, device=dev, dtype=torch.bfloat16)
+    out = _hip().fconv_plain(a.contiguous(), b.contiguous(), _pack(w), bias,
+                             3, 3, 0, buf, 16)
+    assert out.data_ptr() == buf.data_ptr()
+    xr = torch.cat([a, b], dim=-1).float().permute(0, 3, 1, 2)
+    ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias, padding=1)
+    ref = ref.permute(0, 2, 3, 1)
+    got = buf[..., 16:16 + N].float()
+    assert (got - ref).abs().max().item() < 0.05, \
+        (got - ref).abs().max().item()
+    assert (buf[..., :16] == 0).all() and (buf[..., 16 + N:] == 0).all()
+
+
+def test_fconv_small_cin_seam(dev):
+    """Cin=2 (the flow input) exercises the scalar seam path."""
+    B, H, W, N = 1, 7, 9, 32
+    x = torch.randn(B, H, W, 2, device=dev).to(torch.bfloat16)
+    w = torch.randn(N, 2, 7, 7, device=dev) * 0.1
+    bias = torch.randn(N, device=dev)
+    out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, 7, 7,
+                             1, None, 0)
+    ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
+                          w.to(torch.bfloat16).float(), bias, padding=3))
+    ref = ref.permute(0, 2, 3, 1)
+    assert (out.float() - ref).abs().max().item() < 0.05
+
+
+def test_fconv_gru_pair_matches_eager(dev):
+    from raft_amd.models.update import SepConvGRU
+    torch.manual_seed(3)
+    hd, xd = 128, 256
+    B, H, W = 1, 8, 16
+    gru = SepConvGRU(hd, xd).to(dev).to(torch.bfloat16).eval()
+    h = (torch.randn(B, hd, H, W, device=dev) * 0.5).to(torch.bfloat16)
+    x = (torch.randn(B, xd, H, W, device=dev) * 0.5).to(torch.bfloat16)
+    with torch.no_grad():
+        ref = gru(h, x).float()
+    from raft_amd.models.fused import _GruDir
+    hp = h.permute(0, 2, 3, 1).contiguous()
+    xp = x.permute(0, 2, 3, 1).contiguous()
+    d1 = _GruDir(gru.convz1, gru.convr1, gru.convq1)
+    d2 = _GruDir(gru.convz2, gru.convr2, gru.convq2)
+    hip = _hip()
+    out = d2(hip, d1(hip, hp, xp), xp)
+    out = out.permute(0, 3, 1, 2).float()
+    err = (out - ref).abs().max().item()
+    assert err < 0.08, err
+
+
+def test_corr_volume_nhwc_matches_ref(dev):
+    B, H, W, C = 2, 9, 15, 128
+    f = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
+    g = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
+    out = _hip().corr_volume_nhwc(f.contiguous(), g.contiguous(), False)
+    ref = R.corr_volume(f.float().permute(0, 3, 1, 2),
+                        g.float().permute(0, 3, 1, 2))
+    assert out.shape == ref.shape
+    err = (out - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 0.02, err
+
+
+def test_corr_lookup_nhwc_matches_ref(dev):
+    B, H, W, r = 1, 8, 12, 4
+    pyr = [torch.randn(B, H * W, H, W, device=dev)]
+    for _ in range(3):
+        pyr.append(F.avg_pool2d(pyr[-1], 2, 2))
+    coords = torch.rand(B, H, W, 2, device=dev) * 14.0 - 1.0
+    C = 4 * 81
+    cpad = 328
+    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False)
+    assert out.shape == (B, H, W, cpad)
+    ref = R.corr_lookup(pyr, coords, r).permute(0, 2, 3, 1)
+    assert torch.allclose(out[..., :C], ref, atol=1e-4, rtol=1e-4)
+    assert (out[..., C:] == 0).all()
+
+
+def test_fused_model_matches_eager_bf16(dev):
+    from raft_amd import RAFT, RaftConfig
+    import raft_amd.models.fused as fused
+    for small in (False, True):
+        torch.manual_seed(11)
+        m = RAFT(RaftConfig(small=small)).to(dev).eval()
+        x1 = torch.rand(1, 3, 64, 96, device=dev)
+        x2 = torch.rand(1, 3, 64, 96, device=dev)
+        with torch.no_grad():
+            ref32 = m(x1, x2, iters=6)          # fp32 eager (no fuse: fp32)
+        mb = m.to(torch.bfloat16)
+        x1b = x1.to(torch.bfloat16)
+        x2b = x2.to(torch.bfloat16)
+        with torch.no_grad():
+            out = mb(x1b, x2b, iters=6)          # fused path picks up
+        assert fused.can_fuse(mb, x1b)
+        assert out.shape == ref32.shape
+        # recurrent bf16 vs fp32 drift: require close flow fields
+        err = (out.float() - ref32).abs().mean().item()
+        assert err < 0.35, (small, err)
+
+
+def test_fused_cache_invalidates_on_weight_change(dev):
+    from raft_amd import RAFT, RaftConfig
+    import raft_amd.models.fused as fused
+    m = RAFT(RaftConfig(small=True)).to(dev).to(torch.bfloat16).eval()
+    x = torch.rand(1, 3, 64, 96, device=dev, dtype=torch.bfloat16)
+    with torch.no_grad():
+        o1 = m(x, x, iters=2)
+        f1 = m._fused_cache
+        for p in m.update_block.parameters():
+            p.add_(0.01)
+        o2 = m(x, x, iters=2)
+        assert m._fused_cache is not f1       # repacked
+        assert not torch.equal(o1, o2)
