@@ -103,3 +103,52 @@ class InferenceEngine:
         with torch.cuda.graph(graph):
             out = self.model(in1, in2, iters=iters)
         return graph, in1, in2, out
+
+
+class Prefetcher:
+    """Async host->device double-buffering on a dedicated HIP copy stream —
+    the rebuild's equivalent of the reference's
+    ``StagingInput(QueueInput(ds))`` (infer_raft.py:37): the next batch's
+    H2D copy overlaps the current batch's compute.
+
+    Wraps any iterable of (im1, im2) CPU tensor batches; yields device
+    tensors. Pins host memory for true async copies.
+    """
+
+    def __init__(self, iterable, device: torch.device,
+                 dtype: torch.dtype = torch.float32):
+        self.source = iterable
+        self.device = device
+        self.dtype = dtype
+        self.use_cuda = device.type == "cuda"
+        self.copy_stream = torch.cuda.Stream() if self.use_cuda else None
+
+    def __iter__(self):
+        it = iter(self.source)
+        if not self.use_cuda:
+            for im1, im2 in it:
+                yield im1.to(self.device, self.dtype), \
+                    im2.to(self.device, self.dtype)
+            return
+        nxt = self._start_copy(it)
+        while nxt is not None:
+            torch.cuda.current_stream().wait_stream(self.copy_stream)
+            cur = nxt[:2]
+            # the tensors were produced on copy_stream; record them so the
+            # allocator doesn't recycle while in use on the main stream
+            for t in cur:
+                t.record_stream(torch.cuda.current_stream())
+            nxt = self._start_copy(it)
+            yield cur
+
+    def _start_copy(self, it):
+        try:
+            im1, im2 = next(it)
+        except StopIteration:
+            return None
+        with torch.cuda.stream(self.copy_stream):
+            d1 = im1.pin_memory().to(self.device, self.dtype,
+                                     non_blocking=True)
+            d2 = im2.pin_memory().to(self.device, self.dtype,
+                                     non_blocking=True)
+        return d1, d2

@@ -86,6 +86,25 @@ class Trainer:
             cycle_momentum=False, anneal_strategy="linear")
         self.step_count = 0
 
+    # ---------------------------------------------------------- checkpoint
+    def save(self, path: str) -> None:
+        """Full training state (model + optimizer + scheduler + step) —
+        resume capability the reference lacked entirely."""
+        torch.save({
+            "model": self.raw_model.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+            "scheduler": self.scheduler.state_dict(),
+            "step": self.step_count,
+            "cfg": vars(self.cfg),
+        }, path)
+
+    def load(self, path: str) -> None:
+        state = torch.load(path, map_location=self.device, weights_only=True)
+        self.raw_model.load_state_dict(state["model"])
+        self.optimizer.load_state_dict(state["optimizer"])
+        self.scheduler.load_state_dict(state["scheduler"])
+        self.step_count = state["step"]
+
     def step(self, image1: torch.Tensor, image2: torch.Tensor,
              flow_gt: torch.Tensor,
              valid: Optional[torch.Tensor] = None) -> dict:

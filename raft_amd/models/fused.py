@@ -86,7 +86,7 @@ class FusedBasicUpdate:
         enc = ub.encoder
         self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
         self.c2 = _FC(enc.convc2)
-        self.f1 = _FC(enc.convf1)
+        self.f1 = enc.convf1                         # Cin=2: MIOpen NHWC
         self.f2 = _FC(enc.convf2)
         self.cv = _FC(enc.conv)                      # in: [cor(192)|flo(64)]
         self.gru1 = _GruDir(ub.gru.convz1, ub.gru.convr1, ub.gru.convq1)
@@ -100,7 +100,8 @@ class FusedBasicUpdate:
     def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
         # motion encoder (model_utils.py:110-119), outputs into x_buf slice
         cor = self.c2(hip, self.c1(hip, corr_pad))
-        flo = self.f2(hip, self.f1(hip, flow_bf))
+        flo1 = torch.relu(self.f1(flow_bf.permute(0, 3, 1, 2)))
+        flo = self.f2(hip, flo1.permute(0, 2, 3, 1).contiguous())
         ctx = self.ctx_dim
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
         x_buf[..., ctx + 126:ctx + 128] = flow_bf
@@ -119,7 +120,7 @@ class FusedSmallUpdate:
     def __init__(self, ub, corr_cpad: int, ctx_dim: int):
         enc = ub.encoder
         self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
-        self.f1 = _FC(enc.convf1)
+        self.f1 = enc.convf1                         # Cin=2: MIOpen NHWC
         self.f2 = _FC(enc.convf2)
         self.cv = _FC(enc.conv)                      # in: [cor(96)|flo(32)]
         self.gru = _GruDir(ub.gru.convz, ub.gru.convr, ub.gru.convq)
@@ -130,7 +131,8 @@ class FusedSmallUpdate:
 
     def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
         cor = self.c1(hip, corr_pad)
-        flo = self.f2(hip, self.f1(hip, flow_bf))
+        flo1 = torch.relu(self.f1(flow_bf.permute(0, 3, 1, 2)))
+        flo = self.f2(hip, flo1.permute(0, 2, 3, 1).contiguous())
         ctx = self.ctx_dim
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
         x_buf[..., ctx + 80:ctx + 82] = flow_bf
@@ -236,7 +238,11 @@ def get_fused(model) -> Optional[FusedRaft]:
 
 
 def can_fuse(model, image1: torch.Tensor) -> bool:
+    import os
+
     import raft_amd.ops as O
+    if os.environ.get("RAFT_AMD_NO_FUSE", "0") == "1":
+        return False
     try:
         p = next(model.update_block.parameters())
     except StopIteration:

@@ -56,7 +56,10 @@ RAFT_DEV float factivate(float v, int act) {
 #define EP_GRU_ZR 1   // N = 2*hd: [z | r] -> z_buf = sig(z), rh = sig(r)*h
 #define EP_GRU_Q 2    // N = hd: h' = (1-z)*h + z*tanh(q)
 
-template <int KH, int KW>
+// MI/NJ: per-wave 16x16 fragment repeats; tile = (32*MI) x (32*NJ*2)
+// with the fixed 2x2 wave layout. (2,4) = 64x128 (compute-efficient);
+// (1,2) = 32x64 (4x the workgroups — batch-1 grids on 256 CUs).
+template <int KH, int KW, int MI, int NJ>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
     const __hip_bfloat16* __restrict__ in2, int C2,   // may be null/0
@@ -69,28 +72,30 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     __hip_bfloat16* __restrict__ z_buf_out,           // [B,H,W,hd] (EP_GRU_ZR)
     __hip_bfloat16* __restrict__ rh_out) {            // [B,H,W,hd] (EP_GRU_ZR)
     constexpr int TAPS = KH * KW;
-    constexpr int AW = FC_BM + KW - 1;   // A slab width (positions + halo)
+    constexpr int BM = 32 * MI;          // block output positions
+    constexpr int BN = 32 * NJ;          // block output channels
+    constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
     __shared__ char sA[AW * FC_ROWB];
-    __shared__ char sB[KW * FC_BN * FC_ROWB];
+    __shared__ char sB[KW * BN * FC_ROWB];
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
-    const int tiles_per_row = (W + FC_BM - 1) / FC_BM;
+    const int tiles_per_row = (W + BM - 1) / BM;
     const int y = blockIdx.y / tiles_per_row;
-    const int x0 = (blockIdx.y % tiles_per_row) * FC_BM;
-    const int n0 = blockIdx.x * FC_BN;
+    const int x0 = (blockIdx.y % tiles_per_row) * BM;
+    const int n0 = blockIdx.x * BN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
-    const int wm = (wave >> 1) * 32;
-    const int wn = (wave & 1) * 64;
+    const int wm = (wave >> 1) * (16 * MI);
+    const int wn = (wave & 1) * (16 * NJ);
     const long long HW = (long long)H * W;
 
-    floatx4 acc[2][4];
+    floatx4 acc[MI][NJ];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < MI; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
 #pragma unroll
     for (int ty = 0; ty < KH; ++ty) {
@@ -135,10 +140,10 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                 }
                 *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
             }
-            // ---- stage all KW weight tiles [FC_BN][FC_BK]
-            for (int e = tid; e < KW * FC_BN * (FC_BK / 8); e += 256) {
-                const int t = e / (FC_BN * (FC_BK / 8));
-                const int rem = e % (FC_BN * (FC_BK / 8));
+            // ---- stage all KW weight tiles [BN][FC_BK]
+            for (int e = tid; e < KW * BN * (FC_BK / 8); e += 256) {
+                const int t = e / (BN * (FC_BK / 8));
+                const int rem = e % (BN * (FC_BK / 8));
                 const int n = rem / (FC_BK / 8);
                 const int c8 = (rem % (FC_BK / 8)) * 8;
                 uint4v v = {0, 0, 0, 0};
@@ -155,27 +160,27 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                             : (__hip_bfloat16)0.f;
                     v = *(const uint4v*)tmp;
                 }
-                *(uint4v*)(sB + t * FC_BN * FC_ROWB + fswz(n, c8 * 2)) = v;
+                *(uint4v*)(sB + t * BN * FC_ROWB + fswz(n, c8 * 2)) = v;
             }
             __syncthreads();
 
 #pragma unroll
             for (int tx = 0; tx < KW; ++tx) {
-                short8 af[2], bf[4];
+                short8 af[MI], bf[NJ];
                 const unsigned cb = (lane >> 4) * 16;
 #pragma unroll
-                for (int i = 0; i < 2; ++i)
+                for (int i = 0; i < MI; ++i)
                     af[i] = *(const short8*)(
                         sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
 #pragma unroll
-                for (int j = 0; j < 4; ++j)
+                for (int j = 0; j < NJ; ++j)
                     bf[j] = *(const short8*)(
-                        sB + tx * FC_BN * FC_ROWB +
+                        sB + tx * BN * FC_ROWB +
                         fswz(wn + j * 16 + (lane & 15), cb));
 #pragma unroll
-                for (int i = 0; i < 2; ++i)
+                for (int i = 0; i < MI; ++i)
 #pragma unroll
-                    for (int j = 0; j < 4; ++j)
+                    for (int j = 0; j < NJ; ++j)
                         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                             af[i], bf[j], acc[i][j], 0, 0, 0);
             }
@@ -186,9 +191,9 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     // ------------------------------------------------------------- epilogue
     const int hd = (mode == EP_GRU_ZR) ? N / 2 : N;
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < MI; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int m = wm + i * 16 + (lane >> 4) * 4 + r;
@@ -232,13 +237,23 @@ extern "C" void launch_fconv_nhwc_bf16(
     const float* bias, void* out, int B, int H, int W, int N, int n_off,
     int out_cstride, int kh, int kw, int act, int mode, const void* h_state,
     const void* z_buf_in, void* z_buf_out, void* rh_out, hipStream_t s) {
-    const int tiles_per_row = cdiv(W, FC_BM);
-    dim3 grid(cdiv(N, FC_BN), H * tiles_per_row, B);
     dim3 blk(256);
+    // large tile (64x128) when it still fills the chip, else small (32x64):
+    // MI355X has 256 CUs / 8 XCDs — batch-1 grids need the small tile.
+    const long long big_blocks =
+        (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
+    const bool big = big_blocks >= 512;
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
-        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW>), grid, blk, 0, s,     \
-                           FCONV_ARGS);                                      \
+        if (big) {                                                           \
+            dim3 grid(cdiv(N, 128), H * cdiv(W, 64), B);                     \
+            hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, 2, 4>), grid,      \
+                               blk, 0, s, FCONV_ARGS);                       \
+        } else {                                                             \
+            dim3 grid(cdiv(N, 64), H * cdiv(W, 32), B);                      \
+            hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, 1, 2>), grid,      \
+                               blk, 0, s, FCONV_ARGS);                       \
+        }                                                                    \
         return;                                                              \
     }
     FC_CASE(1, 1)

@@ -134,25 +134,29 @@ def test_corr_lookup_nhwc_matches_ref(dev):
 
 
 def test_fused_model_matches_eager_bf16(dev):
+    """Fused NHWC loop vs the eager bf16 path (same dtype, both fp32-accum:
+    only intermediate-rounding differences — measured ~5e-3 max at 6 iters,
+    tools/debug_fused.py)."""
+    import os
+
     from raft_amd import RAFT, RaftConfig
     import raft_amd.models.fused as fused
     for small in (False, True):
         torch.manual_seed(11)
-        m = RAFT(RaftConfig(small=small)).to(dev).eval()
-        x1 = torch.rand(1, 3, 64, 96, device=dev)
-        x2 = torch.rand(1, 3, 64, 96, device=dev)
+        m = RAFT(RaftConfig(small=small)).to(dev).to(torch.bfloat16).eval()
+        x1 = torch.rand(1, 3, 64, 96, device=dev, dtype=torch.bfloat16)
+        x2 = torch.rand(1, 3, 64, 96, device=dev, dtype=torch.bfloat16)
         with torch.no_grad():
-            ref32 = m(x1, x2, iters=6)          # fp32 eager (no fuse: fp32)
-        mb = m.to(torch.bfloat16)
-        x1b = x1.to(torch.bfloat16)
-        x2b = x2.to(torch.bfloat16)
-        with torch.no_grad():
-            out = mb(x1b, x2b, iters=6)          # fused path picks up
-        assert fused.can_fuse(mb, x1b)
-        assert out.shape == ref32.shape
-        # recurrent bf16 vs fp32 drift: require close flow fields
-        err = (out.float() - ref32).abs().mean().item()
-        assert err < 0.35, (small, err)
+            os.environ["RAFT_AMD_NO_FUSE"] = "1"
+            try:
+                ref = m(x1, x2, iters=6)
+            finally:
+                os.environ.pop("RAFT_AMD_NO_FUSE")
+            out = m(x1, x2, iters=6)
+        assert fused.can_fuse(m, x1)
+        assert out.shape == ref.shape
+        err = (out.float() - ref.float()).abs().max().item()
+        assert err < 0.05, (small, err)
 
 
 def test_fused_cache_invalidates_on_weight_change(dev):

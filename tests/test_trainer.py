@@ -41,3 +41,27 @@ def test_trainer_loss_decreases_on_fixed_batch():
     gt = torch.zeros(1, 2, 64, 96)
     losses = [tr.step(x1, x2, gt)["loss"] for _ in range(8)]
     assert losses[-1] < losses[0], losses
+
+
+def test_trainer_checkpoint_resume(tmp_path):
+    torch.manual_seed(1)
+    cfg = TrainConfig(num_steps=6, iters=2, height=64, width=96, batch=1)
+    tr = Trainer(RAFT(RaftConfig(small=True)), cfg,
+                 device=torch.device("cpu"))
+    x1 = torch.rand(1, 3, 64, 96)
+    x2 = torch.rand(1, 3, 64, 96)
+    gt = torch.zeros(1, 2, 64, 96)
+    for _ in range(3):
+        tr.step(x1, x2, gt)
+    p = str(tmp_path / "ckpt.pt")
+    tr.save(p)
+    s_after_4 = tr.step(x1, x2, gt)
+
+    torch.manual_seed(2)  # different init: load must restore everything
+    tr2 = Trainer(RAFT(RaftConfig(small=True)), cfg,
+                  device=torch.device("cpu"))
+    tr2.load(p)
+    assert tr2.step_count == 3
+    s2 = tr2.step(x1, x2, gt)
+    assert abs(s2["loss"] - s_after_4["loss"]) < 1e-5
+    assert abs(s2["lr"] - s_after_4["lr"]) < 1e-9
