@@ -86,7 +86,7 @@ class FusedBasicUpdate:
         enc = ub.encoder
         self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
         self.c2 = _FC(enc.convc2)
-        self.f1 = enc.convf1                         # Cin=2: MIOpen NHWC
+        self.f1 = _FC(enc.convf1)                    # Cin=2: small-K direct
         self.f2 = _FC(enc.convf2)
         self.cv = _FC(enc.conv)                      # in: [cor(192)|flo(64)]
         self.gru1 = _GruDir(ub.gru.convz1, ub.gru.convr1, ub.gru.convq1)
@@ -100,8 +100,9 @@ class FusedBasicUpdate:
     def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
         # motion encoder (model_utils.py:110-119), outputs into x_buf slice
         cor = self.c2(hip, self.c1(hip, corr_pad))
-        flo1 = torch.relu(self.f1(flow_bf.permute(0, 3, 1, 2)))
-        flo = self.f2(hip, flo1.permute(0, 2, 3, 1).contiguous())
+        flo1 = hip.fconv_smallk(flow_bf, self.f1.wp, self.f1.bias,
+                                self.f1.kh, self.f1.kw, ACT_RELU)
+        flo = self.f2(hip, flo1)
         ctx = self.ctx_dim
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
         x_buf[..., ctx + 126:ctx + 128] = flow_bf
@@ -120,7 +121,7 @@ class FusedSmallUpdate:
     def __init__(self, ub, corr_cpad: int, ctx_dim: int):
         enc = ub.encoder
         self.c1 = _FC(enc.convc1, pad_cin=corr_cpad)
-        self.f1 = enc.convf1                         # Cin=2: MIOpen NHWC
+        self.f1 = _FC(enc.convf1)                    # Cin=2: small-K direct
         self.f2 = _FC(enc.convf2)
         self.cv = _FC(enc.conv)                      # in: [cor(96)|flo(32)]
         self.gru = _GruDir(ub.gru.convz, ub.gru.convr, ub.gru.convq)
@@ -131,8 +132,9 @@ class FusedSmallUpdate:
 
     def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
         cor = self.c1(hip, corr_pad)
-        flo1 = torch.relu(self.f1(flow_bf.permute(0, 3, 1, 2)))
-        flo = self.f2(hip, flo1.permute(0, 2, 3, 1).contiguous())
+        flo1 = hip.fconv_smallk(flow_bf, self.f1.wp, self.f1.bias,
+                                self.f1.kh, self.f1.kw, ACT_RELU)
+        flo = self.f2(hip, flo1)
         ctx = self.ctx_dim
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
         x_buf[..., ctx + 80:ctx + 82] = flow_bf
@@ -207,10 +209,13 @@ class FusedRaft:
             coords1 = coords1 + flow_init.permute(0, 2, 3, 1).float()
 
         mask = None
+        # preallocated lookup output: pad channels zeroed once, reused
+        corr_buf = torch.zeros(B, H8, W8, self.corr_cpad, device=net.device,
+                               dtype=torch.bfloat16)
         for _ in range(iters):
             corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
                                             cfg.corr_radius, self.corr_cpad,
-                                            True)
+                                            True, corr_buf)
             flow_bf = (coords1 - coords0).to(torch.bfloat16)
             net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
             coords1 = coords1 + dflow.float()

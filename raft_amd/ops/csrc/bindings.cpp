@@ -46,6 +46,9 @@ void launch_fconv_nhwc_bf16(const void*, int, const void*, int, const void*,
                             const float*, void*, int, int, int, int, int,
                             int, int, int, int, int, const void*,
                             const void*, void*, void*, hipStream_t);
+void launch_fconv_smallk_nhwc_bf16(const void*, const void*, const float*,
+                                   void*, int, int, int, int, int, int,
+                                   int, int, hipStream_t);
 }
 
 namespace {
@@ -230,7 +233,8 @@ at::Tensor corr_pool2x_bf16(at::Tensor corr) {
 
 at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
                             at::Tensor coords, int64_t radius,
-                            int64_t c_stride, bool out_bf16) {
+                            int64_t c_stride, bool out_bf16,
+                            c10::optional<at::Tensor> out_buf) {
     // coords: [B, H, W, 2] fp32; returns physical NHWC [B, H, W, c_stride]
     // with channels [L*KK..c_stride) zero-filled (pad for fconv Cin%8).
     CHECK_DEV(coords); CHECK_CONT(coords);
@@ -253,8 +257,15 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
     // first C channels of every row with the Cs stride).
     auto opts = coords.options().dtype(out_bf16 ? at::kBFloat16
                                                  : at::kFloat);
-    auto out = (c_stride == C) ? at::empty({B, H, W, (int64_t)c_stride}, opts)
-                               : at::zeros({B, H, W, (int64_t)c_stride}, opts);
+    at::Tensor out;
+    if (out_buf.has_value()) {
+        out = out_buf.value();   // caller owns pad-channel zeroing
+        TORCH_CHECK(out.is_contiguous() && out.size(3) == c_stride);
+    } else {
+        out = (c_stride == C)
+            ? at::empty({B, H, W, (int64_t)c_stride}, opts)
+            : at::zeros({B, H, W, (int64_t)c_stride}, opts);
+    }
     launch_corr_lookup_nhwc(ptrs, hs, ws, vol_bf16,
                             coords.data_ptr<float>(), out.data_ptr(),
                             out_bf16, B, H, W, L, (int)radius,
@@ -337,6 +348,21 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
     return out;
 }
 
+at::Tensor fconv_smallk(at::Tensor in1, at::Tensor wp,
+                        c10::optional<at::Tensor> bias, int64_t kh,
+                        int64_t kw, int64_t act) {
+    CHECK_DEV(in1); CHECK_CONT(in1); CHECK_CONT(wp);
+    const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
+    const int C = in1.size(3);
+    const int N = wp.size(1);
+    const float* bptr = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+    auto out = at::empty({B, H, W, N}, in1.options());
+    launch_fconv_smallk_nhwc_bf16(in1.data_ptr(), wp.data_ptr(), bptr,
+                                  out.data_ptr(), B, H, W, C, N, (int)kh,
+                                  (int)kw, (int)act, current_stream());
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -357,5 +383,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fconv_plain", &fconv_plain,
           "fused NHWC bf16 conv (+bias +activation, slice output)");
     m.def("fconv_gru_zr", &fconv_gru_zr, "GRU z/r gate conv pair");
+    m.def("fconv_smallk", &fconv_smallk, "direct NHWC conv for tiny Cin");
     m.def("fconv_gru_q", &fconv_gru_q, "GRU candidate conv + state update");
 }

@@ -265,3 +265,53 @@ extern "C" void launch_fconv_nhwc_bf16(
     FC_CASE(7, 1)
 #undef FC_CASE
 }
+
+// --------------------------------------------------------- small-K direct
+// Direct conv for tiny input-channel counts (the motion encoder's flow
+// branch: convf1 is 7x7 over Cin=2 — model_utils.py:114). K = Cin*kh*kw is
+// far below MFMA efficiency and MIOpen's graph-capture fallback for it is
+// a 106 us CK kernel; this is a plain VALU kernel: one thread per output,
+// windows ride L1 (neighboring n-threads share them), weights L1-resident.
+extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
+    const __hip_bfloat16* __restrict__ in,    // [B, H, W, C], C small
+    const __hip_bfloat16* __restrict__ wp,    // [kh*kw][N][C]
+    const float* __restrict__ bias,           // [N] or null
+    __hip_bfloat16* __restrict__ out,         // [B, H, W, N]
+    int H, int W, int C, int N, int kh, int kw, int act, long long total) {
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         idx < total; idx += stride) {
+        const int n = (int)(idx % N);
+        const long long q = idx / N;
+        const int x = (int)(q % W);
+        const int y = (int)((q / W) % H);
+        const int b = (int)(q / ((long long)W * H));
+        float acc = bias ? bias[n] : 0.0f;
+        for (int ty = 0; ty < kh; ++ty) {
+            const int yy = y + ty - kh / 2;
+            if (yy < 0 || yy >= H) continue;
+            for (int tx = 0; tx < kw; ++tx) {
+                const int xx = x + tx - kw / 2;
+                if (xx < 0 || xx >= W) continue;
+                const __hip_bfloat16* src =
+                    in + (((long long)b * H + yy) * W + xx) * C;
+                const __hip_bfloat16* wr =
+                    wp + ((size_t)(ty * kw + tx) * N + n) * C;
+                for (int c = 0; c < C; ++c)
+                    acc = fmaf((float)src[c], (float)wr[c], acc);
+            }
+        }
+        out[idx] = (__hip_bfloat16)factivate(acc, act);
+    }
+}
+
+extern "C" void launch_fconv_smallk_nhwc_bf16(
+    const void* in, const void* wp, const float* bias, void* out, int B,
+    int H, int W, int C, int N, int kh, int kw, int act, hipStream_t s) {
+    const long long total = (long long)B * H * W * N;
+    int blocks = (int)min((total + 255) / 256, (long long)4096);
+    hipLaunchKernelGGL(fconv_smallk_nhwc_bf16_k, dim3(blocks), dim3(256), 0,
+                       s, (const __hip_bfloat16*)in,
+                       (const __hip_bfloat16*)wp, bias,
+                       (__hip_bfloat16*)out, H, W, C, N, kh, kw, act, total);
+}

@@ -1,0 +1,139 @@
+"""Optical-flow serving: FastAPI app around the InferenceEngine.
+
+The reference had no serving story (single-shot CLI only); this is the
+production-serving layer for the rebuild: HTTP endpoints for flow
+estimation with per-request dynamic shapes (pad8 + HIP-graph shape
+buckets), Prometheus metrics, and health checks.
+
+Run:  python -m raft_amd.serving.server --load w.npz --port 8000
+Test: tests/test_serving.py (CPU, starlette TestClient)
+"""
+from __future__ import annotations
+
+import argparse
+import io
+import struct
+import time
+from typing import Optional
+
+import numpy as np
+import torch
+
+try:
+    from fastapi import FastAPI, Request, Response
+    from prometheus_client import (Counter, Histogram,
+                                   generate_latest, CONTENT_TYPE_LATEST)
+    _HAVE_SERVING = True
+except ImportError:  # pragma: no cover
+    _HAVE_SERVING = False
+
+
+def _decode_png_bytes(data: bytes) -> torch.Tensor:
+    """PNG bytes -> [1,3,H,W] float BGR in [0,1] (via the in-repo codec)."""
+    import tempfile
+
+    from raft_amd.data.imageio import read_png
+    with tempfile.NamedTemporaryFile(suffix=".png") as f:
+        f.write(data)
+        f.flush()
+        img = read_png(f.name)
+    t = torch.from_numpy(img.astype(np.float32) / 255.0)
+    return t.permute(2, 0, 1).unsqueeze(0)
+
+
+def _flo_bytes(flow: np.ndarray) -> bytes:
+    buf = io.BytesIO()
+    buf.write(b"PIEH")
+    buf.write(struct.pack("<ii", flow.shape[1], flow.shape[0]))
+    buf.write(flow.astype(np.float32).tobytes())
+    return buf.getvalue()
+
+
+def create_app(model=None, iters: Optional[int] = None,
+               dtype: torch.dtype = torch.float32):
+    """Build the FastAPI app. model defaults to random-init raft-things."""
+    if not _HAVE_SERVING:
+        raise RuntimeError("fastapi / prometheus_client not installed")
+    from raft_amd import RAFT, RaftConfig
+    from raft_amd.engine.inference import InferenceEngine
+
+    if model is None:
+        model = RAFT(RaftConfig(small=False))
+    engine = InferenceEngine(model, iters=iters, dtype=dtype)
+
+    app = FastAPI(title="raft_amd flow service")
+    requests_total = Counter("raft_requests_total", "flow requests served")
+    request_errors = Counter("raft_request_errors_total", "failed requests")
+    latency = Histogram("raft_request_seconds", "end-to-end request latency")
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok", "device": str(engine.device),
+                "cuda": torch.cuda.is_available()}
+
+    @app.get("/metrics")
+    def metrics():
+        return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+
+    @app.post("/flow")
+    async def flow(request: Request, fmt: str = "flo",
+                   iters: Optional[int] = None):
+        """Body framing (no multipart dependency):
+        [uint32-le len(png1)] [png1 bytes] [png2 bytes]."""
+        t0 = time.perf_counter()
+        try:
+            body = await request.body()
+            (n1,) = struct.unpack_from("<I", body, 0)
+            im1 = _decode_png_bytes(body[4:4 + n1])
+            im2 = _decode_png_bytes(body[4 + n1:])
+            out = engine(im1, im2, iters=iters)
+            flow_np = out[0].float().permute(1, 2, 0).cpu().numpy()
+            if fmt == "color":
+                from raft_amd.utils.flow_viz import flow_to_color
+                from raft_amd.data.imageio import write_png
+                import tempfile
+                color = flow_to_color(flow_np, convert_to_bgr=True)
+                with tempfile.NamedTemporaryFile(suffix=".png") as f:
+                    write_png(f.name, color)
+                    f.seek(0)
+                    payload = open(f.name, "rb").read()
+                media = "image/png"
+            else:
+                payload = _flo_bytes(flow_np)
+                media = "application/octet-stream"
+            requests_total.inc()
+            latency.observe(time.perf_counter() - t0)
+            return Response(payload, media_type=media)
+        except Exception:
+            request_errors.inc()
+            raise
+
+    return app
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser()
+    p.add_argument("--load", default=None)
+    p.add_argument("--small", action="store_true")
+    p.add_argument("--iters", type=int, default=None)
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8000)
+    args = p.parse_args(argv)
+
+    from raft_amd import RAFT, RaftConfig
+    model = RAFT(RaftConfig(small=args.small))
+    if args.load:
+        from raft_amd.utils import checkpoint as ckpt
+        ckpt.load_npz(model, args.load)
+    if torch.cuda.is_available():
+        model = model.to("cuda")
+    app = create_app(model, iters=args.iters,
+                     dtype=torch.bfloat16 if args.dtype == "bf16"
+                     else torch.float32)
+    import uvicorn
+    uvicorn.run(app, host=args.host, port=args.port)
+
+
+if __name__ == "__main__":
+    main()

@@ -126,7 +126,7 @@ def test_corr_lookup_nhwc_matches_ref(dev):
     coords = torch.rand(B, H, W, 2, device=dev) * 14.0 - 1.0
     C = 4 * 81
     cpad = 328
-    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False)
+    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False, None)
     assert out.shape == (B, H, W, cpad)
     ref = R.corr_lookup(pyr, coords, r).permute(0, 2, 3, 1)
     assert torch.allclose(out[..., :C], ref, atol=1e-4, rtol=1e-4)
@@ -152,8 +152,9 @@ def test_fused_model_matches_eager_bf16(dev):
                 ref = m(x1, x2, iters=6)
             finally:
                 os.environ.pop("RAFT_AMD_NO_FUSE")
+            assert fused.can_fuse(m, x1)     # inside no_grad
             out = m(x1, x2, iters=6)
-        assert fused.can_fuse(m, x1)
+        assert getattr(m, "_fused_cache", None) is not None
         assert out.shape == ref.shape
         err = (out.float() - ref.float()).abs().max().item()
         assert err < 0.05, (small, err)
@@ -172,3 +173,15 @@ def test_fused_cache_invalidates_on_weight_change(dev):
         o2 = m(x, x, iters=2)
         assert m._fused_cache is not f1       # repacked
         assert not torch.equal(o1, o2)
+
+
+def test_fconv_smallk_matches_conv2d(dev):
+    B, H, W, C, N = 1, 9, 13, 2, 64
+    x = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
+    w = torch.randn(N, C, 7, 7, device=dev) * 0.1
+    bias = torch.randn(N, device=dev)
+    out = _hip().fconv_smallk(x.contiguous(), _pack(w), bias, 7, 7, 1)
+    ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
+                          w.to(torch.bfloat16).float(), bias, padding=3))
+    ref = ref.permute(0, 2, 3, 1)
+    assert (out.float() - ref).abs().max().item() < 0.02

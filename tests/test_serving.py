@@ -1,0 +1,65 @@
+"""Serving layer tests (CPU, in-process TestClient)."""
+import io
+
+import numpy as np
+import pytest
+import torch
+
+pytest.importorskip("fastapi")
+from starlette.testclient import TestClient
+
+from raft_amd import RAFT, RaftConfig
+from raft_amd.data.imageio import write_png
+from raft_amd.serving.server import create_app
+
+
+@pytest.fixture(scope="module")
+def client(tmp_path_factory):
+    model = RAFT(RaftConfig(small=True)).eval()
+    app = create_app(model, iters=2)
+    return TestClient(app)
+
+
+def _png_bytes(tmp_path, name, h=32, w=48):
+    img = (np.random.rand(h, w, 3) * 255).astype(np.uint8)
+    p = str(tmp_path / name)
+    write_png(p, img)
+    return open(p, "rb").read()
+
+
+def test_healthz(client):
+    r = client.get("/healthz")
+    assert r.status_code == 200
+    assert r.json()["status"] == "ok"
+
+
+def _body(b1, b2):
+    import struct
+    return struct.pack("<I", len(b1)) + b1 + b2
+
+
+def test_flow_flo(client, tmp_path):
+    b1 = _png_bytes(tmp_path, "a.png")
+    b2 = _png_bytes(tmp_path, "b.png")
+    r = client.post("/flow", content=_body(b1, b2))
+    assert r.status_code == 200
+    data = r.content
+    assert data[:4] == b"PIEH"
+    w, h = np.frombuffer(data[4:12], np.int32)
+    assert (w, h) == (48, 32)
+    flow = np.frombuffer(data[12:], np.float32).reshape(h, w, 2)
+    assert np.isfinite(flow).all()
+
+
+def test_flow_color_png(client, tmp_path):
+    b1 = _png_bytes(tmp_path, "c.png")
+    b2 = _png_bytes(tmp_path, "d.png")
+    r = client.post("/flow?fmt=color", content=_body(b1, b2))
+    assert r.status_code == 200
+    assert r.content[:8] == b"\x89PNG\r\n\x1a\n"
+
+
+def test_metrics(client):
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    assert b"raft_requests_total" in r.content
