@@ -187,7 +187,11 @@ class FusedRaft:
         vol = hip.corr_volume_nhwc(f1p, f2p, True)       # bf16 volume
         levels = [vol]
         for _ in range(cfg.corr_levels - 1):
-            levels.append(hip.corr_pool2x_bf16(levels[-1]))
+            last = levels[-1]
+            if last.shape[-2] < 2 or last.shape[-1] < 2:
+                levels.append(last)               # degenerate tiny level
+            else:
+                levels.append(hip.corr_pool2x_bf16(last))
 
         cnet = model.cnet(img1)
         net, inp = torch.split(cnet, [cfg.hidden_dim, cfg.context_dim], dim=1)

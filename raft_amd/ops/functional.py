@@ -91,7 +91,11 @@ def corr_pyramid(fmap1: torch.Tensor, fmap2: torch.Tensor,
     if _use_hip(corr):
         levels = [corr]
         for _ in range(num_levels - 1):
-            levels.append(_CorrPool2x.apply(levels[-1]))
+            last = levels[-1]
+            if last.shape[-2] < 2 or last.shape[-1] < 2:
+                levels.append(last)
+            else:
+                levels.append(_CorrPool2x.apply(last))
         return levels
     return torch_ref.corr_pyramid_pool(corr, num_levels)
 

@@ -38,6 +38,11 @@ def corr_pyramid_pool(corr: torch.Tensor, num_levels: int = 4) -> List[torch.Ten
     """
     pyramid = [corr]
     for _ in range(num_levels - 1):
+        if corr.shape[-2] < 2 or corr.shape[-1] < 2:
+            # degenerate level (tiny input): keep the last level so the
+            # channel count contract (L*KK) holds; lookup clamps anyway
+            pyramid.append(corr)
+            continue
         corr = F.avg_pool2d(corr, 2, stride=2)  # default floor, matches VALID
         pyramid.append(corr)
     return pyramid
