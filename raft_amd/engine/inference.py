@@ -70,6 +70,13 @@ class InferenceEngine:
         if self.device.type == "cuda":
             image1 = image1.contiguous(memory_format=torch.channels_last)
             image2 = image2.contiguous(memory_format=torch.channels_last)
+        from raft_amd.models import fused
+        if fused.can_fuse(self.model, image1):
+            # the fused path captures its own iteration-loop graphs
+            # (MIOpen encoder convs stay eager: capture-time workspace
+            # fallbacks made whole-model graphs slower)
+            self.model._fused_use_graph = self.use_graph
+            return unpad(self.model(image1, image2, iters=iters), hw)
         if not self.use_graph:
             return unpad(self.model(image1, image2, iters=iters), hw)
 

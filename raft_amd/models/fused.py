@@ -163,12 +163,43 @@ class FusedRaft:
                                            cfg.context_dim)
             self.x_dim = cfg.context_dim + 128
         self._version = self._weights_version()
+        self._graphs = {}
 
     def _weights_version(self) -> int:
         return sum(p._version for p in self.model.update_block.parameters())
 
     def stale(self) -> bool:
         return self._version != self._weights_version()
+
+    @torch.no_grad()
+    def _loop(self, levels, net, x_buf, coords0, coords1, iters):
+        """The refinement loop + final upsample — entirely in-repo kernels,
+        so it is hipGraph-capturable (BASELINE config 5) without MIOpen's
+        capture-time workspace fallbacks."""
+        cfg = self.model.cfg
+        hip = self.hip
+        B, H8, W8, _ = coords0.shape
+        corr_buf = torch.zeros(B, H8, W8, self.corr_cpad,
+                               device=net.device, dtype=torch.bfloat16)
+        mask = None
+        for _ in range(iters):
+            corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
+                                            cfg.corr_radius, self.corr_cpad,
+                                            True, corr_buf)
+            flow_bf = (coords1 - coords0).to(torch.bfloat16)
+            net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
+            coords1 = coords1 + dflow.float()
+
+        flow = coords1 - coords0                         # [B,H,W,2] fp32
+        if cfg.small:
+            from raft_amd.ops import torch_ref
+            up = torch_ref.upflow8(flow.permute(0, 3, 1, 2))
+            if cfg.scale_small_upflow:
+                up = 8.0 * up
+            return up
+        flow_nchw = flow.permute(0, 3, 1, 2).contiguous()
+        mask_nchw = mask.permute(0, 3, 1, 2).contiguous()
+        return hip.convex_upsample(flow_nchw, mask_nchw)
 
     @torch.no_grad()
     def run(self, image1, image2, iters, flow_init=None):
@@ -198,10 +229,6 @@ class FusedRaft:
         net = torch.tanh(net).permute(0, 2, 3, 1).contiguous()
         inp = torch.relu(inp).permute(0, 2, 3, 1).contiguous()
 
-        x_buf = torch.empty(B, H8, W8, self.x_dim, device=net.device,
-                            dtype=torch.bfloat16)
-        x_buf[..., :cfg.context_dim] = inp
-
         ys, xs = torch.meshgrid(
             torch.arange(H8, device=net.device, dtype=torch.float32),
             torch.arange(W8, device=net.device, dtype=torch.float32),
@@ -212,29 +239,58 @@ class FusedRaft:
         if flow_init is not None:                        # [B,2,H,W] logical
             coords1 = coords1 + flow_init.permute(0, 2, 3, 1).float()
 
-        mask = None
-        # preallocated lookup output: pad channels zeroed once, reused
-        corr_buf = torch.zeros(B, H8, W8, self.corr_cpad, device=net.device,
-                               dtype=torch.bfloat16)
-        for _ in range(iters):
-            corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
-                                            cfg.corr_radius, self.corr_cpad,
-                                            True, corr_buf)
-            flow_bf = (coords1 - coords0).to(torch.bfloat16)
-            net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
-            coords1 = coords1 + dflow.float()
+        use_graph = getattr(model, "_fused_use_graph", True)
+        if not use_graph:
+            x_buf = torch.empty(B, H8, W8, self.x_dim, device=net.device,
+                                dtype=torch.bfloat16)
+            x_buf[..., :cfg.context_dim] = inp
+            return self._loop(levels, net, x_buf, coords0, coords1, iters)
 
-        flow = (coords1 - coords0)                       # [B,H,W,2] fp32
-        if cfg.small:
-            from raft_amd.ops import torch_ref
-            up = torch_ref.upflow8(flow.permute(0, 3, 1, 2))
-            if cfg.scale_small_upflow:
-                up = 8.0 * up
-            return up
-        flow_nchw = flow.permute(0, 3, 1, 2).contiguous()
-        mask_nchw = mask.permute(0, 3, 1, 2).contiguous()
-        return hip.convex_upsample(flow_nchw, mask_nchw)
+        key = (B, H8, W8, iters)
+        entry = self._graphs.get(key)
+        if entry is None:
+            if len(self._graphs) >= 4:     # shape-bucket cap
+                x_buf = torch.empty(B, H8, W8, self.x_dim,
+                                    device=net.device, dtype=torch.bfloat16)
+                x_buf[..., :cfg.context_dim] = inp
+                return self._loop(levels, net, x_buf, coords0, coords1,
+                                  iters)
+            entry = self._capture(levels, net, inp, coords0, coords1, iters)
+            self._graphs[key] = entry
+        graph, st = entry
+        for dst, src in zip(st["levels"], levels):
+            dst.copy_(src)
+        st["net"].copy_(net)
+        st["x_buf"][..., :cfg.context_dim].copy_(inp)
+        st["coords1"].copy_(coords1)
+        graph.replay()
+        return st["out"]
 
+    def _capture(self, levels, net, inp, coords0, coords1, iters):
+        cfg = self.model.cfg
+        st = {
+            "levels": [l.clone() for l in levels],
+            "net": net.clone(),
+            "x_buf": torch.empty(net.shape[0], net.shape[1], net.shape[2],
+                                 self.x_dim, device=net.device,
+                                 dtype=torch.bfloat16),
+            "coords0": coords0.clone(),
+            "coords1": coords1.clone(),
+        }
+        st["x_buf"][..., :cfg.context_dim] = inp
+        # warm up on a side stream (allocator steady-state before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._loop(st["levels"], st["net"], st["x_buf"],
+                           st["coords0"], st["coords1"], iters)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            st["out"] = self._loop(st["levels"], st["net"], st["x_buf"],
+                                   st["coords0"], st["coords1"], iters)
+        return graph, st
 
 def get_fused(model) -> Optional[FusedRaft]:
     """Return (building if needed) the packed fused runner for this model.

@@ -75,8 +75,11 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int BM = 32 * MI;          // block output positions
     constexpr int BN = 32 * NJ;          // block output channels
     constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
-    __shared__ char sA[AW * FC_ROWB];
-    __shared__ char sB[KW * BN * FC_ROWB];
+    constexpr int ABYTES = AW * FC_ROWB;
+    constexpr int BBYTES = KW * BN * FC_ROWB;
+    // single __shared__ object (guide §5 trap 4a), double-buffered:
+    // [2 x (A slab | KW B tiles)]
+    __shared__ char smem[2 * (ABYTES + BBYTES)];
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
@@ -90,6 +93,8 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const int wm = (wave >> 1) * (16 * MI);
     const int wn = (wave & 1) * (16 * NJ);
     const long long HW = (long long)H * W;
+    const int ksteps = (Cin + FC_BK - 1) / FC_BK;
+    const int nsteps = KH * ksteps;
 
     floatx4 acc[MI][NJ];
 #pragma unroll
@@ -97,95 +102,108 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
 #pragma unroll
         for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-#pragma unroll
-    for (int ty = 0; ty < KH; ++ty) {
+    // stage step s (= ty * ksteps + kstep) into buffer `buf`
+    auto stage = [&](int s, int buf) {
+        char* sA = smem + buf * (ABYTES + BBYTES);
+        char* sB = sA + ABYTES;
+        const int ty = s / ksteps;
+        const int k0 = (s - ty * ksteps) * FC_BK;
         const int row = y + ty - KH / 2;
         const bool row_ok = (row >= 0 && row < H);
-        for (int k0 = 0; k0 < Cin; k0 += FC_BK) {
-            // ---- stage A slab: AW rows x FC_BK ch (8 bf16 per thread pass)
-            for (int e = tid; e < AW * (FC_BK / 8); e += 256) {
-                const int ar = e / (FC_BK / 8);
-                const int c8 = (e % (FC_BK / 8)) * 8;
-                const int x = x0 + ar - KW / 2;
-                uint4v v = {0, 0, 0, 0};
-                if (row_ok && x >= 0 && x < W) {
-                    const int k = k0 + c8;
-                    const long long p = ((long long)b * H + row) * W + x;
-                    if (k < C1) {
-                        if (k + 8 <= C1)
-                            v = *(const uint4v*)(in1 + p * C1 + k);
-                        else {  // straddles the in1|in2 seam: scalar gather
-                            __hip_bfloat16 tmp[8];
-                            for (int u = 0; u < 8; ++u) {
-                                const int kk = k + u;
-                                tmp[u] = (kk < C1) ? in1[p * C1 + kk]
-                                       : (kk - C1 < C2 ? in2[p * C2 + kk - C1]
-                                                       : (__hip_bfloat16)0.f);
-                            }
-                            v = *(const uint4v*)tmp;
+        // ---- A slab: AW rows x FC_BK ch (16 B per thread pass)
+        for (int e = tid; e < AW * (FC_BK / 8); e += 256) {
+            const int ar = e / (FC_BK / 8);
+            const int c8 = (e % (FC_BK / 8)) * 8;
+            const int x = x0 + ar - KW / 2;
+            uint4v v = {0, 0, 0, 0};
+            if (row_ok && x >= 0 && x < W) {
+                const int k = k0 + c8;
+                const long long p = ((long long)b * H + row) * W + x;
+                if (k < C1) {
+                    if (k + 8 <= C1)
+                        v = *(const uint4v*)(in1 + p * C1 + k);
+                    else {  // straddles the in1|in2 seam: scalar gather
+                        __hip_bfloat16 tmp[8];
+                        for (int u = 0; u < 8; ++u) {
+                            const int kk = k + u;
+                            tmp[u] = (kk < C1) ? in1[p * C1 + kk]
+                                   : (kk - C1 < C2 ? in2[p * C2 + kk - C1]
+                                                   : (__hip_bfloat16)0.f);
                         }
-                    } else if (k - C1 < C2) {
-                        if (k - C1 + 8 <= C2)
-                            v = *(const uint4v*)(in2 + p * C2 + (k - C1));
-                        else {
-                            __hip_bfloat16 tmp[8];
-                            for (int u = 0; u < 8; ++u) {
-                                const int kk = k - C1 + u;
-                                tmp[u] = kk < C2 ? in2[p * C2 + kk]
-                                                 : (__hip_bfloat16)0.f;
-                            }
-                            v = *(const uint4v*)tmp;
+                        v = *(const uint4v*)tmp;
+                    }
+                } else if (k - C1 < C2) {
+                    if (k - C1 + 8 <= C2)
+                        v = *(const uint4v*)(in2 + p * C2 + (k - C1));
+                    else {
+                        __hip_bfloat16 tmp[8];
+                        for (int u = 0; u < 8; ++u) {
+                            const int kk = k - C1 + u;
+                            tmp[u] = kk < C2 ? in2[p * C2 + kk]
+                                             : (__hip_bfloat16)0.f;
                         }
+                        v = *(const uint4v*)tmp;
                     }
                 }
-                *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
             }
-            // ---- stage all KW weight tiles [BN][FC_BK]
-            for (int e = tid; e < KW * BN * (FC_BK / 8); e += 256) {
-                const int t = e / (BN * (FC_BK / 8));
-                const int rem = e % (BN * (FC_BK / 8));
-                const int n = rem / (FC_BK / 8);
-                const int c8 = (rem % (FC_BK / 8)) * 8;
-                uint4v v = {0, 0, 0, 0};
-                const int gn = n0 + n;
-                const int k = k0 + c8;
-                if (gn < N && k + 8 <= Cin)
-                    v = *(const uint4v*)(
-                        wp + ((size_t)(ty * KW + t) * N + gn) * Cin + k);
-                else if (gn < N) {
-                    __hip_bfloat16 tmp[8];
-                    for (int u = 0; u < 8; ++u)
-                        tmp[u] = (k + u < Cin)
-                            ? wp[((size_t)(ty * KW + t) * N + gn) * Cin + k + u]
-                            : (__hip_bfloat16)0.f;
-                    v = *(const uint4v*)tmp;
-                }
-                *(uint4v*)(sB + t * BN * FC_ROWB + fswz(n, c8 * 2)) = v;
+            *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
+        }
+        // ---- KW weight tiles [BN][FC_BK]
+        for (int e = tid; e < KW * BN * (FC_BK / 8); e += 256) {
+            const int t = e / (BN * (FC_BK / 8));
+            const int rem = e % (BN * (FC_BK / 8));
+            const int n = rem / (FC_BK / 8);
+            const int c8 = (rem % (FC_BK / 8)) * 8;
+            uint4v v = {0, 0, 0, 0};
+            const int gn = n0 + n;
+            const int k = k0 + c8;
+            if (gn < N && k + 8 <= Cin)
+                v = *(const uint4v*)(
+                    wp + ((size_t)(ty * KW + t) * N + gn) * Cin + k);
+            else if (gn < N) {
+                __hip_bfloat16 tmp[8];
+                for (int u = 0; u < 8; ++u)
+                    tmp[u] = (k + u < Cin)
+                        ? wp[((size_t)(ty * KW + t) * N + gn) * Cin + k + u]
+                        : (__hip_bfloat16)0.f;
+                v = *(const uint4v*)tmp;
             }
-            __syncthreads();
+            *(uint4v*)(sB + t * BN * FC_ROWB + fswz(n, c8 * 2)) = v;
+        }
+    };
 
+    // 2-phase pipeline (guide §5.5 T3 minimum form): stage(s+1) issues
+    // BEFORE the ds_read+MFMA of step s; ONE barrier per step covers both
+    // "next buffer written" and "current buffer free to overwrite".
+    stage(0, 0);
+    __syncthreads();
+    int cur = 0;
+    for (int s = 0; s < nsteps; ++s) {
+        if (s + 1 < nsteps) stage(s + 1, cur ^ 1);
+        const char* sA = smem + cur * (ABYTES + BBYTES);
+        const char* sB = sA + ABYTES;
 #pragma unroll
-            for (int tx = 0; tx < KW; ++tx) {
-                short8 af[MI], bf[NJ];
-                const unsigned cb = (lane >> 4) * 16;
+        for (int tx = 0; tx < KW; ++tx) {
+            short8 af[MI], bf[NJ];
+            const unsigned cb = (lane >> 4) * 16;
 #pragma unroll
-                for (int i = 0; i < MI; ++i)
-                    af[i] = *(const short8*)(
-                        sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+            for (int i = 0; i < MI; ++i)
+                af[i] = *(const short8*)(
+                    sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+#pragma unroll
+            for (int j = 0; j < NJ; ++j)
+                bf[j] = *(const short8*)(
+                    sB + tx * BN * FC_ROWB +
+                    fswz(wn + j * 16 + (lane & 15), cb));
+#pragma unroll
+            for (int i = 0; i < MI; ++i)
 #pragma unroll
                 for (int j = 0; j < NJ; ++j)
-                    bf[j] = *(const short8*)(
-                        sB + tx * BN * FC_ROWB +
-                        fswz(wn + j * 16 + (lane & 15), cb));
-#pragma unroll
-                for (int i = 0; i < MI; ++i)
-#pragma unroll
-                    for (int j = 0; j < NJ; ++j)
-                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            af[i], bf[j], acc[i][j], 0, 0, 0);
-            }
-            __syncthreads();
+                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        af[i], bf[j], acc[i][j], 0, 0, 0);
         }
+        __syncthreads();
+        cur ^= 1;
     }
 
     // ------------------------------------------------------------- epilogue
@@ -260,9 +278,6 @@ extern "C" void launch_fconv_nhwc_bf16(
     FC_CASE(3, 3)
     FC_CASE(1, 5)
     FC_CASE(5, 1)
-    FC_CASE(7, 7)
-    FC_CASE(1, 7)
-    FC_CASE(7, 1)
 #undef FC_CASE
 }
 

@@ -185,3 +185,22 @@ def test_fconv_smallk_matches_conv2d(dev):
                           w.to(torch.bfloat16).float(), bias, padding=3))
     ref = ref.permute(0, 2, 3, 1)
     assert (out.float() - ref).abs().max().item() < 0.02
+
+
+def test_full_res_volume_config4(dev):
+    """BASELINE config 4: 2x1080x1920, full-res correlation volume resident
+    in HBM (bf16 level-0 is ~2.1 GB at 135x240 queries). Short iter count —
+    this validates memory headroom + non-square tiling, not throughput."""
+    from raft_amd import RAFT, RaftConfig
+    m = RAFT(RaftConfig(small=False)).to(dev).to(torch.bfloat16).eval()
+    x1 = torch.rand(1, 3, 1080, 1920, device=dev, dtype=torch.bfloat16)
+    x2 = torch.rand(1, 3, 1080, 1920, device=dev, dtype=torch.bfloat16)
+    m._fused_use_graph = False     # one-shot: skip capture
+    with torch.no_grad():
+        out = m(x1, x2, iters=2)
+    torch.cuda.synchronize()
+    assert out.shape == (1, 2, 1080, 1920)
+    assert torch.isfinite(out.float()).all()
+    stats = torch.cuda.memory_stats()
+    peak = stats["allocated_bytes.all.peak"] / 2**30
+    assert peak < 40, f"unexpected memory blowup: {peak:.1f} GiB"
