@@ -43,7 +43,9 @@ class InferenceEngine:
 
     def __init__(self, model, iters: Optional[int] = None,
                  dtype: torch.dtype = torch.float32,
-                 use_graph: bool = True, max_graphs: int = 8):
+                 use_graph: bool = True, max_graphs: int = 8,
+                 loop_graph: bool = False):
+        self.loop_graph = loop_graph
         self.model = model.eval()
         self.iters = iters
         self.dtype = dtype
@@ -72,10 +74,10 @@ class InferenceEngine:
             image2 = image2.contiguous(memory_format=torch.channels_last)
         from raft_amd.models import fused
         if fused.can_fuse(self.model, image1):
-            # the fused path captures its own iteration-loop graphs
-            # (MIOpen encoder convs stay eager: capture-time workspace
-            # fallbacks made whole-model graphs slower)
-            self.model._fused_use_graph = self.use_graph
+            # fused path: loop-level graphs exist but default OFF (replay
+            # measured slower than eager on this stack); opt in with
+            # loop_graph=True
+            self.model._fused_use_graph = self.loop_graph
             return unpad(self.model(image1, image2, iters=iters), hw)
         if not self.use_graph:
             return unpad(self.model(image1, image2, iters=iters), hw)

@@ -239,7 +239,11 @@ class FusedRaft:
         if flow_init is not None:                        # [B,2,H,W] logical
             coords1 = coords1 + flow_init.permute(0, 2, 3, 1).float()
 
-        use_graph = getattr(model, "_fused_use_graph", True)
+        # loop-graph replay measured SLOWER than eager on ROCm 7.2 (24.3 vs
+        # 18.6 ms/step): replay overhead exceeds the launch overhead saved
+        # once the loop is ~15 kernels/iter. Capability kept for config 5 /
+        # future stacks; enable per-model via _fused_use_graph = True.
+        use_graph = getattr(model, "_fused_use_graph", False)
         if not use_graph:
             x_buf = torch.empty(B, H8, W8, self.x_dim, device=net.device,
                                 dtype=torch.bfloat16)

@@ -307,6 +307,10 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
         cstride = N;
         n_off = 0;
     }
+    TORCH_CHECK((kh == 1 && kw == 1) || (kh == 3 && kw == 3) ||
+                (kh == 1 && kw == 5) || (kh == 5 && kw == 1),
+                "fconv: unsupported kernel shape ", kh, "x", kw,
+                " (use fconv_smallk for tiny-Cin large kernels)");
     launch_fconv_nhwc_bf16(in1.data_ptr(), C1, p2, C2, wp.data_ptr(), bptr,
                            out.data_ptr(), B, H, W, N, (int)n_off, cstride,
                            (int)kh, (int)kw, (int)act, 0, nullptr, nullptr,

@@ -77,9 +77,10 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
     constexpr int ABYTES = AW * FC_ROWB;
     constexpr int BBYTES = KW * BN * FC_ROWB;
-    // single __shared__ object (guide §5 trap 4a), double-buffered:
-    // [2 x (A slab | KW B tiles)]
-    __shared__ char smem[2 * (ABYTES + BBYTES)];
+    // single-buffered LDS: a 2-phase double buffer was measured SLOWER here
+    // (it halves blocks/CU at these batch-1 grids; occupancy is the
+    // latency-hiding lever, not intra-block pipelining)
+    __shared__ char smem[ABYTES + BBYTES];
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
@@ -172,15 +173,10 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
         }
     };
 
-    // 2-phase pipeline (guide §5.5 T3 minimum form): stage(s+1) issues
-    // BEFORE the ds_read+MFMA of step s; ONE barrier per step covers both
-    // "next buffer written" and "current buffer free to overwrite".
-    stage(0, 0);
-    __syncthreads();
-    int cur = 0;
     for (int s = 0; s < nsteps; ++s) {
-        if (s + 1 < nsteps) stage(s + 1, cur ^ 1);
-        const char* sA = smem + cur * (ABYTES + BBYTES);
+        stage(s, 0);
+        __syncthreads();
+        const char* sA = smem;
         const char* sB = sA + ABYTES;
 #pragma unroll
         for (int tx = 0; tx < KW; ++tx) {
@@ -203,7 +199,6 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                         af[i], bf[j], acc[i][j], 0, 0, 0);
         }
         __syncthreads();
-        cur ^= 1;
     }
 
     // ------------------------------------------------------------- epilogue
