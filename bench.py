@@ -73,6 +73,8 @@ def main():
         args.iters, args.dtype = 12, "fp32"
 
     dev = torch.device(f"cuda:{local_rank}" if have_gpu else "cpu")
+    if have_gpu:
+        torch.backends.cudnn.benchmark = True
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
 
     torch.manual_seed(1234 + rank)

@@ -57,6 +57,11 @@ class InferenceEngine:
             # batched_transpose kernels) and the fused NHWC hot loop gets
             # zero-copy physical views.
             self.model = self.model.to(memory_format=torch.channels_last)
+            # MIOpen's heuristic (immediate) mode sometimes picks
+            # no-workspace CK fallback solvers (~200 us/conv on the
+            # encoders); benchmark mode runs a real find with workspace
+            # during warmup and caches the fast igemm solutions.
+            torch.backends.cudnn.benchmark = True
         self.use_graph = use_graph and self.device.type == "cuda"
         self.max_graphs = max_graphs
         self._graphs: Dict[Tuple[int, int, int], tuple] = {}
