@@ -149,10 +149,12 @@ def main():
 
 def _make_infer_step(model, x1, x2, args):
     iters = args.iters
-    if not args.no_graph and torch.cuda.is_available():
+    if torch.cuda.is_available():
         from raft_amd.engine.inference import InferenceEngine
         engine = InferenceEngine(model, iters=iters, dtype=x1.dtype,
-                                 use_graph=True)
+                                 use_graph=not args.no_graph,
+                                 loop_graph=not args.no_graph and
+                                 os.environ.get("RAFT_AMD_LOOP_GRAPH") == "1")
 
         def step():
             return engine(x1, x2)
