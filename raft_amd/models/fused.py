@@ -63,9 +63,10 @@ class _FC:
     def __init__(self, conv, pad_cin=None, scale=1.0):
         self.wp, self.bias, self.kh, self.kw = pack_conv(conv, pad_cin, scale)
 
-    def __call__(self, hip, in1, in2=None, act=ACT_RELU, out=None, n_off=0):
+    def __call__(self, hip, in1, in2=None, act=ACT_RELU, out=None, n_off=0,
+                 in1_off=0, in1_len=0):
         return hip.fconv_plain(in1, in2, self.wp, self.bias, self.kh,
-                               self.kw, act, out, n_off)
+                               self.kw, act, out, n_off, in1_off, in1_len)
 
 
 class _GruDir:
@@ -91,9 +92,14 @@ class FusedBasicUpdate:
         self.cv = _FC(enc.conv)                      # in: [cor(192)|flo(64)]
         self.gru1 = _GruDir(ub.gru.convz1, ub.gru.convr1, ub.gru.convq1)
         self.gru2 = _GruDir(ub.gru.convz2, ub.gru.convr2, ub.gru.convq2)
+        # flow_head.conv1 and mask[0] are both 3x3(128->256) on net: run as
+        # ONE N=512 conv; their consumers read strided slices of the result
         self.fh1 = _FC(ub.flow_head.conv1)
-        self.fh2 = _FC(ub.flow_head.conv2)
         self.m0 = _FC(ub.mask[0])
+        self.heads_w = torch.cat([self.fh1.wp, self.m0.wp], dim=1) \
+            .contiguous()
+        self.heads_b = torch.cat([self.fh1.bias, self.m0.bias]).contiguous()
+        self.fh2 = _FC(ub.flow_head.conv2)
         self.m2 = _FC(ub.mask[2], scale=0.25)        # fold the 0.25 scale
         self.ctx_dim = ctx_dim
 
@@ -109,9 +115,11 @@ class FusedBasicUpdate:
         # SepConvGRU (model_utils.py:138-156)
         net = self.gru1(hip, net, x_buf)
         net = self.gru2(hip, net, x_buf)
-        # heads
-        dflow = self.fh2(hip, self.fh1(hip, net), act=ACT_NONE)
-        mask = self.m2(hip, self.m0(hip, net), act=ACT_NONE)
+        # heads: one merged 3x3 conv, consumers read strided slices
+        hbuf = hip.fconv_plain(net, None, self.heads_w, self.heads_b, 3, 3,
+                               ACT_RELU, None, 0, 0, 0)
+        dflow = self.fh2(hip, hbuf, act=ACT_NONE, in1_off=0, in1_len=256)
+        mask = self.m2(hip, hbuf, act=ACT_NONE, in1_off=256, in1_len=256)
         return net, mask, dflow
 
 
@@ -181,12 +189,15 @@ class FusedRaft:
         B, H8, W8, _ = coords0.shape
         corr_buf = torch.zeros(B, H8, W8, self.corr_cpad,
                                device=net.device, dtype=torch.bfloat16)
+        flow_buf = torch.empty(B, H8, W8, 2, device=net.device,
+                               dtype=torch.bfloat16)
         mask = None
         for _ in range(iters):
+            # lookup also emits flow = coords1 - identity grid (fused)
             corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
                                             cfg.corr_radius, self.corr_cpad,
-                                            True, corr_buf)
-            flow_bf = (coords1 - coords0).to(torch.bfloat16)
+                                            True, corr_buf, flow_buf)
+            flow_bf = flow_buf
             net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
             coords1 = coords1 + dflow.float()
 

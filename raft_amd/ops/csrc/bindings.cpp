@@ -35,16 +35,16 @@ void launch_convex_upsample_bwd_f32(const float*, const float*, const float*,
 void launch_corr_volume_nhwc_bf16(const void*, const void*, void*, bool,
                                   int, int, int, int, float, hipStream_t);
 void launch_corr_lookup_nhwc(const void* const*, const int*, const int*,
-                             bool, const float*, void*, bool, int, int, int,
-                             int, int, int, hipStream_t);
+                             bool, const float*, void*, bool, void*, int,
+                             int, int, int, int, int, hipStream_t);
 void launch_corr_lookup_nhwc_bwd(float* const*, const int*, const int*,
                                  const float*, const float*, int, int, int,
                                  int, int, hipStream_t);
 void launch_corr_pool2x_bf16(const void*, void*, int, int, int, int,
                              long long, hipStream_t);
-void launch_fconv_nhwc_bf16(const void*, int, const void*, int, const void*,
-                            const float*, void*, int, int, int, int, int,
-                            int, int, int, int, int, const void*,
+void launch_fconv_nhwc_bf16(const void*, int, int, int, const void*, int,
+                            const void*, const float*, void*, int, int, int,
+                            int, int, int, int, int, int, int, const void*,
                             const void*, void*, void*, hipStream_t);
 void launch_fconv_smallk_nhwc_bf16(const void*, const void*, const float*,
                                    void*, int, int, int, int, int, int,
@@ -234,7 +234,8 @@ at::Tensor corr_pool2x_bf16(at::Tensor corr) {
 at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
                             at::Tensor coords, int64_t radius,
                             int64_t c_stride, bool out_bf16,
-                            c10::optional<at::Tensor> out_buf) {
+                            c10::optional<at::Tensor> out_buf,
+                            c10::optional<at::Tensor> flow_buf) {
     // coords: [B, H, W, 2] fp32; returns physical NHWC [B, H, W, c_stride]
     // with channels [L*KK..c_stride) zero-filled (pad for fconv Cin%8).
     CHECK_DEV(coords); CHECK_CONT(coords);
@@ -266,9 +267,15 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
             ? at::empty({B, H, W, (int64_t)c_stride}, opts)
             : at::zeros({B, H, W, (int64_t)c_stride}, opts);
     }
+    void* fptr = nullptr;
+    if (flow_buf.has_value()) {
+        TORCH_CHECK(flow_buf->is_contiguous() &&
+                    flow_buf->scalar_type() == at::kBFloat16);
+        fptr = flow_buf->data_ptr();
+    }
     launch_corr_lookup_nhwc(ptrs, hs, ws, vol_bf16,
                             coords.data_ptr<float>(), out.data_ptr(),
-                            out_bf16, B, H, W, L, (int)radius,
+                            out_bf16, fptr, B, H, W, L, (int)radius,
                             (int)c_stride, current_stream());
     return out;
 }
@@ -276,11 +283,14 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
 at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
                        at::Tensor wp, c10::optional<at::Tensor> bias,
                        int64_t kh, int64_t kw, int64_t act,
-                       c10::optional<at::Tensor> out_buf, int64_t n_off) {
+                       c10::optional<at::Tensor> out_buf, int64_t n_off,
+                       int64_t in1_off, int64_t in1_len) {
     CHECK_DEV(in1); CHECK_CONT(in1); CHECK_DEV(wp); CHECK_CONT(wp);
     TORCH_CHECK(in1.scalar_type() == at::kBFloat16, "fconv needs bf16");
     const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
-    const int C1 = in1.size(3);
+    const int in1_stride = in1.size(3);
+    const int C1 = in1_len > 0 ? (int)in1_len : in1_stride;
+    TORCH_CHECK(in1_off + C1 <= in1_stride);
     const void* p2 = nullptr;
     int C2 = 0;
     if (in2.has_value()) {
@@ -311,10 +321,11 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
                 (kh == 1 && kw == 5) || (kh == 5 && kw == 1),
                 "fconv: unsupported kernel shape ", kh, "x", kw,
                 " (use fconv_smallk for tiny-Cin large kernels)");
-    launch_fconv_nhwc_bf16(in1.data_ptr(), C1, p2, C2, wp.data_ptr(), bptr,
-                           out.data_ptr(), B, H, W, N, (int)n_off, cstride,
-                           (int)kh, (int)kw, (int)act, 0, nullptr, nullptr,
-                           nullptr, nullptr, current_stream());
+    launch_fconv_nhwc_bf16(in1.data_ptr(), C1, in1_stride, (int)in1_off,
+                           p2, C2, wp.data_ptr(), bptr, out.data_ptr(), B,
+                           H, W, N, (int)n_off, cstride, (int)kh, (int)kw,
+                           (int)act, 0, nullptr, nullptr, nullptr, nullptr,
+                           current_stream());
     return out;
 }
 
@@ -327,10 +338,10 @@ std::vector<at::Tensor> fconv_gru_zr(at::Tensor h, at::Tensor x,
     TORCH_CHECK(N == 2 * hd, "zr weights must stack [Wz; Wr]");
     auto z = at::empty_like(h);
     auto rh = at::empty_like(h);
-    launch_fconv_nhwc_bf16(h.data_ptr(), hd, x.data_ptr(), x.size(3),
-                           wp.data_ptr(), bias.data_ptr<float>(), nullptr,
-                           B, H, W, N, 0, 0, (int)kh, (int)kw, 0, 1,
-                           h.data_ptr(), nullptr, z.data_ptr(),
+    launch_fconv_nhwc_bf16(h.data_ptr(), hd, hd, 0, x.data_ptr(),
+                           x.size(3), wp.data_ptr(), bias.data_ptr<float>(),
+                           nullptr, B, H, W, N, 0, 0, (int)kh, (int)kw, 0,
+                           1, h.data_ptr(), nullptr, z.data_ptr(),
                            rh.data_ptr(), current_stream());
     return {z, rh};
 }
@@ -344,11 +355,12 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
     const int hd = rh.size(3);
     TORCH_CHECK(wp.size(1) == hd);
     auto out = at::empty_like(h);
-    launch_fconv_nhwc_bf16(rh.data_ptr(), hd, x.data_ptr(), x.size(3),
-                           wp.data_ptr(), bias.data_ptr<float>(),
-                           out.data_ptr(), B, H, W, hd, 0, hd, (int)kh,
-                           (int)kw, 0, 2, h.data_ptr(), z.data_ptr(),
-                           nullptr, nullptr, current_stream());
+    launch_fconv_nhwc_bf16(rh.data_ptr(), hd, hd, 0, x.data_ptr(),
+                           x.size(3), wp.data_ptr(),
+                           bias.data_ptr<float>(), out.data_ptr(), B, H, W,
+                           hd, 0, hd, (int)kh, (int)kw, 0, 2, h.data_ptr(),
+                           z.data_ptr(), nullptr, nullptr,
+                           current_stream());
     return out;
 }
 

@@ -146,6 +146,7 @@ template <typename T, typename OT>
 __global__ void corr_lookup_nhwc_k(
     LevelsT lv, const float* __restrict__ coords,  // [B, H, W, 2]
     OT* __restrict__ out,                          // [B, H, W, Cs]
+    __hip_bfloat16* __restrict__ flow_out,         // [B, H, W, 2] or null
     int H, int W, int num_levels, int radius, int Cs, long long total) {
     const int K = 2 * radius + 1;
     const int KK = K * K;
@@ -157,6 +158,14 @@ __global__ void corr_lookup_nhwc_k(
         const long long q = idx / C;               // b*H*W + y*W + x
         const int lvl = c / KK;
         const int k = c - lvl * KK;
+
+        if (flow_out && c < 2) {
+            // fused flow = coords - identity grid (coords0 is the pixel
+            // grid by construction, RAFT.py:111-117)
+            const float base = (c == 0) ? (float)(q % W)
+                                        : (float)((q / W) % H);
+            flow_out[q * 2 + c] = (__hip_bfloat16)(coords[q * 2 + c] - base);
+        }
 
         const float inv = 1.0f / (float)(1 << lvl);
         const float cx = coords[q * 2] * inv + (float)(k / K - radius);
@@ -175,8 +184,9 @@ __global__ void corr_lookup_nhwc_k(
 
 extern "C" void launch_corr_lookup_nhwc(
     const void* const* level_ptrs, const int* level_h, const int* level_w,
-    bool vol_bf16, const float* coords, void* out, bool out_bf16, int B,
-    int H, int W, int num_levels, int radius, int Cs, hipStream_t s) {
+    bool vol_bf16, const float* coords, void* out, bool out_bf16,
+    void* flow_out, int B, int H, int W, int num_levels, int radius, int Cs,
+    hipStream_t s) {
     LevelsT lv{};
     for (int i = 0; i < num_levels; ++i) {
         lv.ptr[i] = level_ptrs[i];
@@ -190,7 +200,8 @@ extern "C" void launch_corr_lookup_nhwc(
     if (vol_bf16 == std::is_same<T, __hip_bfloat16>::value &&               \
         out_bf16 == OTC) {                                                  \
         hipLaunchKernelGGL((corr_lookup_nhwc_k<T, OT>), dim3(blocks),       \
-                           dim3(256), 0, s, lv, coords, (OT*)out, H, W,     \
+                           dim3(256), 0, s, lv, coords, (OT*)out,           \
+                           (__hip_bfloat16*)flow_out, H, W,                 \
                            num_levels, radius, Cs, total);                  \
         return;                                                             \
     }

@@ -62,6 +62,7 @@ RAFT_DEV float factivate(float v, int act) {
 template <int KH, int KW, int MI, int NJ>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
+    int in1_stride, int in1_off,                      // strided slice of in1
     const __hip_bfloat16* __restrict__ in2, int C2,   // may be null/0
     const __hip_bfloat16* __restrict__ wp,            // [KH*KW][N][C1+C2]
     const float* __restrict__ bias,                   // [N] or null
@@ -122,12 +123,14 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                 const long long p = ((long long)b * H + row) * W + x;
                 if (k < C1) {
                     if (k + 8 <= C1)
-                        v = *(const uint4v*)(in1 + p * C1 + k);
+                        v = *(const uint4v*)(in1 + p * in1_stride + in1_off
+                                             + k);
                     else {  // straddles the in1|in2 seam: scalar gather
                         __hip_bfloat16 tmp[8];
                         for (int u = 0; u < 8; ++u) {
                             const int kk = k + u;
-                            tmp[u] = (kk < C1) ? in1[p * C1 + kk]
+                            tmp[u] = (kk < C1)
+                                   ? in1[p * in1_stride + in1_off + kk]
                                    : (kk - C1 < C2 ? in2[p * C2 + kk - C1]
                                                    : (__hip_bfloat16)0.f);
                         }
@@ -239,17 +242,19 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
 }
 
 #define FCONV_ARGS                                                           \
-    (const __hip_bfloat16*)in1, C1, (const __hip_bfloat16*)in2, C2,          \
+    (const __hip_bfloat16*)in1, C1, in1_stride, in1_off,                     \
+    (const __hip_bfloat16*)in2, C2,                                          \
     (const __hip_bfloat16*)wp, bias, (__hip_bfloat16*)out, H, W, N, n_off,   \
     out_cstride, act, mode, (const __hip_bfloat16*)h_state,                  \
     (const __hip_bfloat16*)z_buf_in, (__hip_bfloat16*)z_buf_out,             \
     (__hip_bfloat16*)rh_out
 
 extern "C" void launch_fconv_nhwc_bf16(
-    const void* in1, int C1, const void* in2, int C2, const void* wp,
-    const float* bias, void* out, int B, int H, int W, int N, int n_off,
-    int out_cstride, int kh, int kw, int act, int mode, const void* h_state,
-    const void* z_buf_in, void* z_buf_out, void* rh_out, hipStream_t s) {
+    const void* in1, int C1, int in1_stride, int in1_off, const void* in2,
+    int C2, const void* wp, const float* bias, void* out, int B, int H,
+    int W, int N, int n_off, int out_cstride, int kh, int kw, int act,
+    int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
+    void* rh_out, hipStream_t s) {
     dim3 blk(256);
     // large tile (64x128) when it still fills the chip, else small (32x64):
     // MI355X has 256 CUs / 8 XCDs — batch-1 grids need the small tile.
@@ -279,14 +284,71 @@ extern "C" void launch_fconv_nhwc_bf16(
 // --------------------------------------------------------- small-K direct
 // Direct conv for tiny input-channel counts (the motion encoder's flow
 // branch: convf1 is 7x7 over Cin=2 — model_utils.py:114). K = Cin*kh*kw is
-// far below MFMA efficiency and MIOpen's graph-capture fallback for it is
-// a 106 us CK kernel; this is a plain VALU kernel: one thread per output,
-// windows ride L1 (neighboring n-threads share them), weights L1-resident.
-extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
-    const __hip_bfloat16* __restrict__ in,    // [B, H, W, C], C small
+// far below MFMA efficiency. LDS-staged form: weights staged once per
+// block, PPB = 256/N positions per block share them; all-lane-uniform
+// input reads broadcast from LDS.
+#define SK_MAXW (49 * 128 * 4)       // taps x N x C cap (bf16)
+#define SK_MAXIN (8 * 49 * 4)        // PPB x taps x C cap
+
+extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
+    const __hip_bfloat16* __restrict__ in,    // [B, H, W, C]
     const __hip_bfloat16* __restrict__ wp,    // [kh*kw][N][C]
-    const float* __restrict__ bias,           // [N] or null
+    const float* __restrict__ bias,
     __hip_bfloat16* __restrict__ out,         // [B, H, W, N]
+    int H, int W, int C, int N, int kh, int kw, int act, long long ncells) {
+    __shared__ __hip_bfloat16 smem[SK_MAXW + SK_MAXIN];
+    __hip_bfloat16* sw = smem;
+    __hip_bfloat16* sin = smem + SK_MAXW;
+    const int taps = kh * kw;
+    const int PPB = 256 / N;
+    const int tid = threadIdx.x;
+    for (int e = tid; e < taps * N * C; e += 256) sw[e] = wp[e];
+
+    const long long groups = (ncells + PPB - 1) / PPB;
+    for (long long g = blockIdx.x; g < groups; g += gridDim.x) {
+        const long long cell0 = g * (long long)PPB;
+        // stage PPB windows (zero-padded SAME)
+        for (int e = tid; e < PPB * taps * C; e += 256) {
+            const int p = e / (taps * C);
+            const int rem = e % (taps * C);
+            const int t = rem / C;
+            const int c = rem % C;
+            const long long cell = cell0 + p;
+            __hip_bfloat16 v = (__hip_bfloat16)0.f;
+            if (cell < ncells) {
+                const int x = (int)(cell % W);
+                const int y = (int)((cell / W) % H);
+                const int b = (int)(cell / ((long long)W * H));
+                const int yy = y + t / kw - kh / 2;
+                const int xx = x + t % kw - kw / 2;
+                if (yy >= 0 && yy < H && xx >= 0 && xx < W)
+                    v = in[(((long long)b * H + yy) * W + xx) * C + c];
+            }
+            sin[e] = v;
+        }
+        __syncthreads();
+        const int p = tid / N;
+        const int n = tid - p * N;
+        const long long cell = cell0 + p;
+        if (cell < ncells) {
+            float acc = bias ? bias[n] : 0.0f;
+            const __hip_bfloat16* win = sin + p * taps * C;
+            for (int t = 0; t < taps; ++t)
+                for (int c = 0; c < C; ++c)
+                    acc = fmaf((float)win[t * C + c],
+                               (float)sw[(t * N + n) * C + c], acc);
+            out[cell * N + n] = (__hip_bfloat16)factivate(acc, act);
+        }
+        __syncthreads();
+    }
+}
+
+// grid-stride naive fallback for shapes outside the LDS caps
+extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
+    const __hip_bfloat16* __restrict__ in,
+    const __hip_bfloat16* __restrict__ wp,
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,
     int H, int W, int C, int N, int kh, int kw, int act, long long total) {
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -318,7 +380,21 @@ extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
 extern "C" void launch_fconv_smallk_nhwc_bf16(
     const void* in, const void* wp, const float* bias, void* out, int B,
     int H, int W, int C, int N, int kh, int kw, int act, hipStream_t s) {
-    const long long total = (long long)B * H * W * N;
+    const long long ncells = (long long)B * H * W;
+    const int taps = kh * kw;
+    if (N <= 128 && 256 % N == 0 && taps * N * C <= SK_MAXW &&
+        (256 / N) * taps * C <= SK_MAXIN) {
+        const int PPB = 256 / N;
+        const long long groups = (ncells + PPB - 1) / PPB;
+        int blocks = (int)min(groups, (long long)4096);
+        hipLaunchKernelGGL(fconv_smallk_lds_k, dim3(blocks), dim3(256), 0,
+                           s, (const __hip_bfloat16*)in,
+                           (const __hip_bfloat16*)wp, bias,
+                           (__hip_bfloat16*)out, H, W, C, N, kh, kw, act,
+                           ncells);
+        return;
+    }
+    const long long total = ncells * N;
     int blocks = (int)min((total + 255) / 256, (long long)4096);
     hipLaunchKernelGGL(fconv_smallk_nhwc_bf16_k, dim3(blocks), dim3(256), 0,
                        s, (const __hip_bfloat16*)in,

@@ -40,7 +40,7 @@ def test_fconv_matches_conv2d(dev, kh, kw):
     w = torch.randn(N, Cin, kh, kw, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, kh, kw,
-                             1, None, 0)
+                             1, None, 0, 0, 0)
     # reference: fp32 conv on the bf16-rounded inputs/weights
     xr = x.float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias,
@@ -59,7 +59,7 @@ def test_fconv_two_inputs_and_slice_output(dev):
     bias = torch.zeros(N, device=dev)
     buf = torch.zeros(B, H, W, 80, device=dev, dtype=torch.bfloat16)
     out = _hip().fconv_plain(a.contiguous(), b.contiguous(), _pack(w), bias,
-                             3, 3, 0, buf, 16)
+                             3, 3, 0, buf, 16, 0, 0)
     assert out.data_ptr() == buf.data_ptr()
     xr = torch.cat([a, b], dim=-1).float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias, padding=1)
@@ -77,7 +77,7 @@ def test_fconv_small_cin_seam(dev):
     w = torch.randn(N, 12, 3, 3, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, 3, 3,
-                             1, None, 0)
+                             1, None, 0, 0, 0)
     ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=1))
     ref = ref.permute(0, 2, 3, 1)
@@ -126,7 +126,7 @@ def test_corr_lookup_nhwc_matches_ref(dev):
     coords = torch.rand(B, H, W, 2, device=dev) * 14.0 - 1.0
     C = 4 * 81
     cpad = 328
-    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False, None)
+    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False, None, None)
     assert out.shape == (B, H, W, cpad)
     ref = R.corr_lookup(pyr, coords, r).permute(0, 2, 3, 1)
     assert torch.allclose(out[..., :C], ref, atol=1e-4, rtol=1e-4)
@@ -204,3 +204,34 @@ def test_full_res_volume_config4(dev):
     stats = torch.cuda.memory_stats()
     peak = stats["allocated_bytes.all.peak"] / 2**30
     assert peak < 40, f"unexpected memory blowup: {peak:.1f} GiB"
+
+
+def test_fconv_strided_input_slice(dev):
+    """in1_off/in1_len: consume a channel slice of a wider NHWC buffer."""
+    B, H, W = 1, 6, 9
+    buf = torch.randn(B, H, W, 96, device=dev).to(torch.bfloat16)
+    w = torch.randn(24, 32, 3, 3, device=dev) * 0.1
+    bias = torch.zeros(24, device=dev)
+    out = _hip().fconv_plain(buf.contiguous(), None, _pack(w), bias, 3, 3,
+                             0, None, 0, 40, 32)
+    ref = F.conv2d(buf[..., 40:72].float().permute(0, 3, 1, 2),
+                   w.to(torch.bfloat16).float(), bias, padding=1)
+    ref = ref.permute(0, 2, 3, 1)
+    assert (out.float() - ref).abs().max().item() < 0.05
+
+
+def test_lookup_fused_flow_output(dev):
+    B, H, W, r = 1, 6, 8, 2
+    pyr = [torch.randn(B, H * W, H, W, device=dev)]
+    for _ in range(2):
+        pyr.append(F.avg_pool2d(pyr[-1], 2, 2))
+    coords = torch.rand(B, H, W, 2, device=dev) * 8.0
+    flow_buf = torch.empty(B, H, W, 2, device=dev, dtype=torch.bfloat16)
+    C = 3 * 25
+    _hip().corr_lookup_nhwc(list(pyr), coords, r, C, False, None, flow_buf)
+    ys, xs = torch.meshgrid(torch.arange(H, device=dev, dtype=torch.float32),
+                            torch.arange(W, device=dev, dtype=torch.float32),
+                            indexing="ij")
+    grid = torch.stack([xs, ys], dim=-1)[None]
+    ref = (coords - grid).to(torch.bfloat16)
+    assert torch.equal(flow_buf, ref)
