@@ -35,13 +35,18 @@ typedef unsigned int uint4v __attribute__((ext_vector_type(4)));
 #define FC_BM 64
 #define FC_BN 128
 #define FC_BK 32
-// +16 B row pad -> 80-B stride: 16 consecutive rows hit 16 distinct 16-B
-// bank slots mod the 256-B bank row, so 16-lane ds_read_b128 groups (one
-// row per lane, same column) are conflict-free (guide §2/§6 Guideline 4).
-#define FC_ROWB (FC_BK * 2 + 16)
+#define FC_ROWB (FC_BK * 2)   // 64 B rows, swizzled (no pad)
 
+// Grouping-independent XOR swizzle: for any 16 rows at a fixed 16-B column
+// the swizzled addresses hit 16 DISTINCT bank slots (bits 4-7 of the
+// address enumerate (row&3, (row&3)^((row>>2)&3)) — a bijection), so
+// ds_read_b128 is conflict-free regardless of how the HW partitions the 64
+// lanes into service groups. PMC showed ~1100 conflict-cycles/wave with a
+// +16B-pad scheme whose residue argument assumed contiguous lane groups.
+// Bijective per 1-KiB window: bits 4-5 ^= row&3, bits 6-7 ^= (row>>2)&3.
 RAFT_DEV unsigned fswz(int row, unsigned colbyte) {
-    return row * FC_ROWB + colbyte;
+    return (unsigned)(row * FC_ROWB + colbyte) ^
+           (((unsigned)row & 3u) << 4) ^ ((((unsigned)row >> 2) & 3u) << 6);
 }
 
 RAFT_DEV float factivate(float v, int act) {
@@ -59,7 +64,11 @@ RAFT_DEV float factivate(float v, int act) {
 // MI/NJ: per-wave 16x16 fragment repeats; tile = (32*MI) x (32*NJ*2)
 // with the fixed 2x2 wave layout. (2,4) = 64x128 (compute-efficient);
 // (1,2) = 32x64 (4x the workgroups — batch-1 grids on 256 CUs).
-template <int KH, int KW, int MI, int NJ>
+// AT (all-taps staging): stage every kernel-row slab and every tap's
+// weight tile per K-step, so each barrier pair covers KH*KW MFMA groups
+// instead of KW — trades LDS (occupancy) for barrier amortization; only
+// meaningful for KH > 1 (the 5x1 GRU conv had 2 MFMA per barrier).
+template <int KH, int KW, int MI, int NJ, bool AT>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
     int in1_stride, int in1_off,                      // strided slice of in1
@@ -76,12 +85,15 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int BM = 32 * MI;          // block output positions
     constexpr int BN = 32 * NJ;          // block output channels
     constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
-    constexpr int ABYTES = AW * FC_ROWB;
-    constexpr int BBYTES = KW * BN * FC_ROWB;
+    constexpr int APAD = ((AW + 15) / 16) * 16;  // swizzle window rounding
+    constexpr int NSLAB = AT ? KH : 1;
+    constexpr int NBT = AT ? TAPS : KW;
+    constexpr int ABYTES = APAD * FC_ROWB;
+    constexpr int BBYTES = BN * FC_ROWB;
     // single-buffered LDS: a 2-phase double buffer was measured SLOWER here
     // (it halves blocks/CU at these batch-1 grids; occupancy is the
     // latency-hiding lever, not intra-block pipelining)
-    __shared__ char smem[ABYTES + BBYTES];
+    __shared__ char smem[NSLAB * ABYTES + NBT * BBYTES];
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
@@ -96,7 +108,7 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const int wn = (wave & 1) * (16 * NJ);
     const long long HW = (long long)H * W;
     const int ksteps = (Cin + FC_BK - 1) / FC_BK;
-    const int nsteps = KH * ksteps;
+    const int nsteps = AT ? ksteps : KH * ksteps;
 
     floatx4 acc[MI][NJ];
 #pragma unroll
@@ -104,18 +116,23 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
 #pragma unroll
         for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    // stage step s (= ty * ksteps + kstep) into buffer `buf`
-    auto stage = [&](int s, int buf) {
-        char* sA = smem + buf * (ABYTES + BBYTES);
-        char* sB = sA + ABYTES;
-        const int ty = s / ksteps;
-        const int k0 = (s - ty * ksteps) * FC_BK;
-        const int row = y + ty - KH / 2;
-        const bool row_ok = (row >= 0 && row < H);
-        // ---- A slab: AW rows x FC_BK ch (16 B per thread pass)
-        for (int e = tid; e < AW * (FC_BK / 8); e += 256) {
-            const int ar = e / (FC_BK / 8);
-            const int c8 = (e % (FC_BK / 8)) * 8;
+    char* const sAbase = smem;
+    char* const sBbase = smem + NSLAB * ABYTES;
+
+    // stage step s into LDS. !AT: s = ty * ksteps + kstep (one row slab +
+    // KW weight tiles). AT: s = kstep (KH row slabs + TAPS weight tiles).
+    auto stage = [&](int s) {
+        const int ty0 = AT ? 0 : s / ksteps;
+        const int k0 = (AT ? s : (s - ty0 * ksteps)) * FC_BK;
+        // ---- A slabs: NSLAB x AW rows x FC_BK ch (16 B per thread pass)
+        for (int e = tid; e < NSLAB * AW * (FC_BK / 8); e += 256) {
+            const int sl = e / (AW * (FC_BK / 8));
+            const int rem0 = e % (AW * (FC_BK / 8));
+            const int ar = rem0 / (FC_BK / 8);
+            const int c8 = (rem0 % (FC_BK / 8)) * 8;
+            char* sA = sAbase + sl * ABYTES;
+            const int row = y + (ty0 + sl) - KH / 2;
+            const bool row_ok = (row >= 0 && row < H);
             const int x = x0 + ar - KW / 2;
             uint4v v = {0, 0, 0, 0};
             if (row_ok && x >= 0 && x < W) {
@@ -152,8 +169,8 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             }
             *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
         }
-        // ---- KW weight tiles [BN][FC_BK]
-        for (int e = tid; e < KW * BN * (FC_BK / 8); e += 256) {
+        // ---- weight tiles [NBT][BN][FC_BK]
+        for (int e = tid; e < NBT * BN * (FC_BK / 8); e += 256) {
             const int t = e / (BN * (FC_BK / 8));
             const int rem = e % (BN * (FC_BK / 8));
             const int n = rem / (FC_BK / 8);
@@ -161,45 +178,48 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             uint4v v = {0, 0, 0, 0};
             const int gn = n0 + n;
             const int k = k0 + c8;
+            const int tap = AT ? t : (ty0 * KW + t);
             if (gn < N && k + 8 <= Cin)
                 v = *(const uint4v*)(
-                    wp + ((size_t)(ty * KW + t) * N + gn) * Cin + k);
+                    wp + ((size_t)tap * N + gn) * Cin + k);
             else if (gn < N) {
                 __hip_bfloat16 tmp[8];
                 for (int u = 0; u < 8; ++u)
                     tmp[u] = (k + u < Cin)
-                        ? wp[((size_t)(ty * KW + t) * N + gn) * Cin + k + u]
+                        ? wp[((size_t)tap * N + gn) * Cin + k + u]
                         : (__hip_bfloat16)0.f;
                 v = *(const uint4v*)tmp;
             }
-            *(uint4v*)(sB + t * BN * FC_ROWB + fswz(n, c8 * 2)) = v;
+            *(uint4v*)(sBbase + t * BBYTES + fswz(n, c8 * 2)) = v;
         }
     };
 
     for (int s = 0; s < nsteps; ++s) {
-        stage(s, 0);
+        stage(s);
         __syncthreads();
-        const char* sA = smem;
-        const char* sB = sA + ABYTES;
+        const unsigned cb = (lane >> 4) * 16;
 #pragma unroll
-        for (int tx = 0; tx < KW; ++tx) {
-            short8 af[MI], bf[NJ];
-            const unsigned cb = (lane >> 4) * 16;
+        for (int sl = 0; sl < NSLAB; ++sl) {
+            const char* sA = sAbase + sl * ABYTES;
 #pragma unroll
-            for (int i = 0; i < MI; ++i)
-                af[i] = *(const short8*)(
-                    sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+            for (int tx = 0; tx < KW; ++tx) {
+                short8 af[MI], bf[NJ];
 #pragma unroll
-            for (int j = 0; j < NJ; ++j)
-                bf[j] = *(const short8*)(
-                    sB + tx * BN * FC_ROWB +
-                    fswz(wn + j * 16 + (lane & 15), cb));
-#pragma unroll
-            for (int i = 0; i < MI; ++i)
+                for (int i = 0; i < MI; ++i)
+                    af[i] = *(const short8*)(
+                        sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
 #pragma unroll
                 for (int j = 0; j < NJ; ++j)
-                    acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        af[i], bf[j], acc[i][j], 0, 0, 0);
+                    bf[j] = *(const short8*)(
+                        sBbase + (sl * KW + tx) * BBYTES +
+                        fswz(wn + j * 16 + (lane & 15), cb));
+#pragma unroll
+                for (int i = 0; i < MI; ++i)
+#pragma unroll
+                    for (int j = 0; j < NJ; ++j)
+                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            af[i], bf[j], acc[i][j], 0, 0, 0);
+            }
         }
         __syncthreads();
     }
@@ -254,31 +274,37 @@ extern "C" void launch_fconv_nhwc_bf16(
     int C2, const void* wp, const float* bias, void* out, int B, int H,
     int W, int N, int n_off, int out_cstride, int kh, int kw, int act,
     int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
-    void* rh_out, hipStream_t s) {
+    void* rh_out, int alltaps, hipStream_t s) {
     dim3 blk(256);
+    // alltaps: -1 auto (KH>1 && KW==1 -> on), 0 off, 1 on
+    const bool at = (alltaps < 0) ? (kh > 1 && kw == 1) : (alltaps != 0);
     // large tile (64x128) when it still fills the chip, else small (32x64):
     // MI355X has 256 CUs / 8 XCDs — batch-1 grids need the small tile.
     const long long big_blocks =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
     const bool big = big_blocks >= 512;
+#define FC_LAUNCH(KH, KW, MI, NJ, AT, BMv, BNv)                              \
+    {                                                                        \
+        dim3 grid(cdiv(N, BNv), H * cdiv(W, BMv), B);                        \
+        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, MI, NJ, AT>), grid,    \
+                           blk, 0, s, FCONV_ARGS);                           \
+        return;                                                              \
+    }
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
-        if (big) {                                                           \
-            dim3 grid(cdiv(N, 128), H * cdiv(W, 64), B);                     \
-            hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, 2, 4>), grid,      \
-                               blk, 0, s, FCONV_ARGS);                       \
-        } else {                                                             \
-            dim3 grid(cdiv(N, 64), H * cdiv(W, 32), B);                      \
-            hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, 1, 2>), grid,      \
-                               blk, 0, s, FCONV_ARGS);                       \
+        if (KH > 1 && at) {                                                  \
+            if (big) FC_LAUNCH(KH, KW, 2, 4, true, 64, 128)                  \
+            else FC_LAUNCH(KH, KW, 1, 2, true, 32, 64)                       \
         }                                                                    \
-        return;                                                              \
+        if (big) FC_LAUNCH(KH, KW, 2, 4, false, 64, 128)                     \
+        else FC_LAUNCH(KH, KW, 1, 2, false, 32, 64)                          \
     }
     FC_CASE(1, 1)
     FC_CASE(3, 3)
     FC_CASE(1, 5)
     FC_CASE(5, 1)
 #undef FC_CASE
+#undef FC_LAUNCH
 }
 
 // --------------------------------------------------------- small-K direct
