@@ -261,6 +261,58 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             }
 }
 
+// ----------------------------------------------------------- tiny-N conv
+// N <= 4 (the flow head's final 3x3 -> 2): an MFMA tile wastes 97% of its
+// columns. One wave per position: lanes stride over K = taps*Cin with N
+// accumulators each, then a cross-lane tree reduce per n.
+template <int NN>
+__global__ __launch_bounds__(256) void fconv_tinyn_k(
+    const __hip_bfloat16* __restrict__ in,    // [B,H,W,Cin] (full rows)
+    int Cin, int in_stride, int in_off,
+    const __hip_bfloat16* __restrict__ wp,    // [taps][N][Cin]
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,         // [B,H,W,N]
+    int H, int W, int kh, int kw, int act, long long ncells) {
+    const int lane = threadIdx.x & 63;
+    const long long cell = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (cell >= ncells) return;
+    const int x = (int)(cell % W);
+    const int y = (int)((cell / W) % H);
+    const int b = (int)(cell / ((long long)W * H));
+    const int taps = kh * kw;
+    const int K = taps * Cin;
+
+    float acc[NN];
+#pragma unroll
+    for (int n = 0; n < NN; ++n) acc[n] = 0.0f;
+
+    for (int k = lane; k < K; k += 64) {
+        const int t = k / Cin;
+        const int c = k - t * Cin;
+        const int yy = y + t / kw - kh / 2;
+        const int xx = x + t % kw - kw / 2;
+        if (yy < 0 || yy >= H || xx < 0 || xx >= W) continue;
+        const float v = (float)in[(((long long)b * H + yy) * W + xx)
+                                  * in_stride + in_off + c];
+#pragma unroll
+        for (int n = 0; n < NN; ++n)
+            acc[n] = fmaf(v, (float)wp[((size_t)t * NN + n) * Cin + c],
+                          acc[n]);
+    }
+#pragma unroll
+    for (int n = 0; n < NN; ++n)
+        for (int off = 32; off > 0; off >>= 1)
+            acc[n] += __shfl_down(acc[n], off, 64);
+    if (lane == 0) {    // tree reduce leaves each total in lane 0
+#pragma unroll
+        for (int n = 0; n < NN; ++n) {
+            float v = acc[n] + (bias ? bias[n] : 0.0f);
+            out[cell * NN + n] = (__hip_bfloat16)factivate(v, act);
+        }
+    }
+}
+
+
 #define FCONV_ARGS                                                           \
     (const __hip_bfloat16*)in1, C1, in1_stride, in1_off,                     \
     (const __hip_bfloat16*)in2, C2,                                          \
@@ -276,13 +328,31 @@ extern "C" void launch_fconv_nhwc_bf16(
     int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
     void* rh_out, int alltaps, hipStream_t s) {
     dim3 blk(256);
-    // alltaps: -1 auto (KH>1 && KW==1 -> on), 0 off, 1 on
-    const bool at = (alltaps < 0) ? (kh > 1 && kw == 1) : (alltaps != 0);
+    const long long big_blocks0 =
+        (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
+    const bool big0 = big_blocks0 >= 512;
+    // alltaps (measured, tools/bench_fconv.py): wins for kh>1 on the small
+    // tile (5x1: 43.6 -> 32.8 us; 3x3 convc2: 35.5 -> 28.4), loses on the
+    // big tile (heads 3x3 N=512: 30.2 -> 39.8; LDS kills occupancy)
+    const bool at = (alltaps < 0) ? (kh > 1 && !big0) : (alltaps != 0);
     // large tile (64x128) when it still fills the chip, else small (32x64):
     // MI355X has 256 CUs / 8 XCDs — batch-1 grids need the small tile.
     const long long big_blocks =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
     const bool big = big_blocks >= 512;
+    if (N <= 4 && mode == 0 && in2 == nullptr && n_off == 0 &&
+        out_cstride == N) {
+        const long long ncells = (long long)B * H * W;
+        dim3 tg((unsigned)((ncells + 3) / 4));
+        if (N == 2) {
+            hipLaunchKernelGGL(fconv_tinyn_k<2>, tg, blk, 0, s,
+                               (const __hip_bfloat16*)in1, C1, in1_stride,
+                               in1_off, (const __hip_bfloat16*)wp, bias,
+                               (__hip_bfloat16*)out, H, W, kh, kw, act,
+                               ncells);
+            return;
+        }
+    }
 #define FC_LAUNCH(KH, KW, MI, NJ, AT, BMv, BNv)                              \
     {                                                                        \
         dim3 grid(cdiv(N, BNv), H * cdiv(W, BMv), B);                        \

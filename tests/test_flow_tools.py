@@ -1,0 +1,79 @@
+"""Flow post-processing extras (reference flow_utils.py beyond the RAFT
+path): reversal+hole-fill, static masking, USM aug, guided filter."""
+import numpy as np
+import torch
+
+from raft_amd.utils.flow_tools import (aug_img, box_filter, calc_flow,
+                                       gaussian_blur, guided_filter,
+                                       reverse_flow, set_static_flow)
+
+
+def test_reverse_flow_uniform_shift():
+    """A uniform +3px x-shift reverses to a uniform -3px shift."""
+    h, w = 16, 24
+    flow = np.zeros((h, w, 2), np.float32)
+    flow[:, :, 0] = 3.0
+    rev, holes = reverse_flow(flow)
+    # interior pixels (reachable targets) must be exactly -3
+    assert np.allclose(rev[:, 5:w - 1, 0], -3.0)
+    assert np.allclose(rev[:, :, 1], 0.0)
+    # the 3 left columns are unreachable -> holes filled from neighbors
+    assert holes[:, :3].all()
+    assert np.allclose(rev[:, :3, 0], -3.0)   # hole fill propagates -3
+
+
+def test_reverse_flow_conflict_averaging():
+    """Two sources landing on one target average their (negated) flows."""
+    flow = np.zeros((1, 4, 2), np.float32)
+    flow[0, 0, 0] = 2.0   # 0 -> 2
+    flow[0, 1, 0] = 1.0   # 1 -> 2
+    flow[0, 2, 0] = 3.0   # 2 -> out of range, clipped to 3
+    rev, _ = reverse_flow(flow)
+    assert abs(rev[0, 2, 0] + 1.5) < 1e-6     # avg(-2, -1)
+
+
+def test_set_static_flow():
+    im0 = np.full((4, 4, 3), 100, np.uint8)
+    bg = im0.copy()
+    bg[0, 0] = 200                            # one moving pixel
+    flow = np.ones((4, 4, 2), np.float32)
+    out = set_static_flow(flow, im0, bg)
+    assert out[0, 0, 0] == 1.0
+    assert (out[1:] == 0).all()
+
+
+def test_aug_img_shapes_and_range():
+    im = (np.random.rand(20, 30, 3) * 255).astype(np.uint8)
+    out = aug_img(im)
+    assert out.shape == im.shape and out.dtype == np.uint8
+
+
+def test_gaussian_blur_preserves_constant():
+    im = np.full((16, 16), 7.0)
+    assert np.allclose(gaussian_blur(im, 2.0), 7.0)
+
+
+def test_box_filter_mean():
+    im = np.arange(25, dtype=np.float64).reshape(5, 5)
+    out = box_filter(im, 1)
+    assert abs(out[2, 2] - im[1:4, 1:4].mean()) < 1e-9
+
+
+def test_guided_filter_smooths_noise():
+    rng = np.random.default_rng(0)
+    guide = np.zeros((32, 32), np.float64)
+    guide[:, 16:] = 255.0                     # step edge
+    src = guide / 255.0 + rng.normal(0, 0.2, (32, 32))
+    out = guided_filter(guide, src, radius=4, eps=1e-2)
+    # noise reduced away from the edge; edge preserved
+    assert out[:, :8].std() < src[:, :8].std() * 0.5
+    assert abs(out[:, 24:].mean() - 1.0) < 0.15
+
+
+def test_calc_flow_runs_model():
+    torch.manual_seed(0)
+    im0 = (np.random.rand(64, 96, 3) * 255).astype(np.uint8)
+    im1 = np.roll(im0, 2, axis=1)
+    flow = calc_flow(im0, im1, iters=2, post_filter=True)
+    assert flow.shape == (64, 96, 2)
+    assert np.isfinite(flow).all()
