@@ -55,6 +55,7 @@ class TrainConfig:
     batch: int = 2
     height: int = 368
     width: int = 768
+    amp: bool = True                  # bf16 autocast on GPU (fp32 grads)
 
 
 class Trainer:
@@ -68,6 +69,10 @@ class Trainer:
             torch.device("cuda") if torch.cuda.is_available()
             else torch.device("cpu"))
         model = model.to(self.device)
+        if self.device.type == "cuda":
+            # channels-last: MIOpen NHWC igemm without transposes
+            model = model.to(memory_format=torch.channels_last)
+            torch.backends.cudnn.benchmark = True
         self.raw_model = model
         if dist.is_available() and dist.is_initialized() and \
                 dist.get_world_size() > 1:
@@ -109,8 +114,17 @@ class Trainer:
              flow_gt: torch.Tensor,
              valid: Optional[torch.Tensor] = None) -> dict:
         self.optimizer.zero_grad(set_to_none=True)
-        preds = self.model(image1, image2, iters=self.cfg.iters,
-                           test_mode=False)
+        use_amp = self.cfg.amp and self.device.type == "cuda"
+        image1 = image1.to(self.device)
+        image2 = image2.to(self.device)
+        flow_gt = flow_gt.to(self.device)
+        if self.device.type == "cuda":
+            image1 = image1.contiguous(memory_format=torch.channels_last)
+            image2 = image2.contiguous(memory_format=torch.channels_last)
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_amp):
+            preds = self.model(image1, image2, iters=self.cfg.iters,
+                               test_mode=False)
         loss = sequence_loss(preds, flow_gt, self.cfg.gamma, valid)
         loss.backward()
         if self.distributed:

@@ -280,24 +280,26 @@ __global__ __launch_bounds__(256) void fconv_tinyn_k(
     const int y = (int)((cell / W) % H);
     const int b = (int)(cell / ((long long)W * H));
     const int taps = kh * kw;
-    const int K = taps * Cin;
 
     float acc[NN];
 #pragma unroll
     for (int n = 0; n < NN; ++n) acc[n] = 0.0f;
 
-    for (int k = lane; k < K; k += 64) {
-        const int t = k / Cin;
-        const int c = k - t * Cin;
+    // tap-outer, channel-inner: one bounds check per tap, lane-contiguous
+    // (coalesced) channel loads
+    for (int t = 0; t < taps; ++t) {
         const int yy = y + t / kw - kh / 2;
         const int xx = x + t % kw - kw / 2;
         if (yy < 0 || yy >= H || xx < 0 || xx >= W) continue;
-        const float v = (float)in[(((long long)b * H + yy) * W + xx)
-                                  * in_stride + in_off + c];
+        const __hip_bfloat16* src =
+            in + (((long long)b * H + yy) * W + xx) * in_stride + in_off;
+        const __hip_bfloat16* wr = wp + (size_t)t * NN * Cin;
+        for (int c = lane; c < Cin; c += 64) {
+            const float v = (float)src[c];
 #pragma unroll
-        for (int n = 0; n < NN; ++n)
-            acc[n] = fmaf(v, (float)wp[((size_t)t * NN + n) * Cin + c],
-                          acc[n]);
+            for (int n = 0; n < NN; ++n)
+                acc[n] = fmaf(v, (float)wr[n * Cin + c], acc[n]);
+        }
     }
 #pragma unroll
     for (int n = 0; n < NN; ++n)

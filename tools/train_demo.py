@@ -1,0 +1,59 @@
+"""End-to-end learning demonstration: train raft-small on synthetic warped
+pairs until EPE drops well below the random-init level. Exercises the full
+training stack (autograd through the HIP lookup/GRU/upsample kernels on
+GPU, sequence loss, AdamW/one-cycle, AMP).
+
+Run: python tools/train_demo.py [--steps 300] [--out curve.json]
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=300)
+    ap.add_argument("--batch", type=int, default=4)
+    ap.add_argument("--size", default="192x320")
+    ap.add_argument("--iters", type=int, default=8)
+    ap.add_argument("--out", default=None)
+    args = ap.parse_args()
+    h, w = (int(v) for v in args.size.split("x"))
+
+    from raft_amd import RAFT, RaftConfig
+    from raft_amd.data.synthetic import synthetic_pair
+    from raft_amd.engine.trainer import Trainer, TrainConfig
+
+    torch.manual_seed(0)
+    cfg = TrainConfig(num_steps=args.steps, iters=args.iters, lr=2e-4,
+                      batch=args.batch, height=h, width=w)
+    tr = Trainer(RAFT(RaftConfig(small=True)), cfg)
+    curve = []
+    t0 = time.perf_counter()
+    for step in range(args.steps):
+        im1, im2, gt = synthetic_pair(args.batch, h, w, seed=step,
+                                      max_mag=8.0)
+        stats = tr.step(im1, im2, gt)
+        if step % 25 == 0 or step == args.steps - 1:
+            curve.append({"step": step, **stats})
+            print(f"step {step:4d}  loss {stats['loss']:.3f}  "
+                  f"epe {stats['epe']:.3f}  lr {stats['lr']:.2e}", flush=True)
+    wall = time.perf_counter() - t0
+    result = {"curve": curve, "wall_s": wall,
+              "ms_per_step": 1e3 * wall / args.steps,
+              "epe_first": curve[0]["epe"], "epe_last": curve[-1]["epe"]}
+    print(json.dumps({k: v for k, v in result.items() if k != "curve"}))
+    if args.out:
+        with open(args.out, "w") as f:
+            json.dump(result, f, indent=2)
+    assert curve[-1]["epe"] < 0.7 * curve[0]["epe"], "no learning progress"
+
+
+if __name__ == "__main__":
+    main()
