@@ -37,6 +37,8 @@ def parse_args():
                    help="bench the training step (config 3) instead")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph capture of the inference step")
+    p.add_argument("--trace", default=None,
+                   help="write a chrome trace of a few steps to this path")
     return p.parse_args()
 
 
@@ -100,6 +102,15 @@ def main():
     for _ in range(args.warmup):
         run_step()
     sync()
+    if args.trace and rank == 0:
+        from torch.profiler import profile, ProfilerActivity
+        with profile(activities=[ProfilerActivity.CPU,
+                                 ProfilerActivity.CUDA]) as prof:
+            for _ in range(3):
+                run_step()
+            sync()
+        prof.export_chrome_trace(args.trace)
+        print(f"trace written to {args.trace}", file=sys.stderr)
     t0 = time.perf_counter()
     for _ in range(args.steps):
         run_step()
@@ -170,30 +181,23 @@ def _make_infer_step(model, x1, x2, args):
 
 
 def _make_train_step(model, args, dev, dist):
-    """Config 3: training step, sequence loss, AdamW, DP grad all-reduce."""
-    from raft_amd.engine.trainer import sequence_loss
-    if dist is not None:
-        from raft_amd.parallel.ddp import BucketedDDP
-        model = BucketedDDP(model)
-    opt = torch.optim.AdamW(model.parameters(), lr=4e-4, weight_decay=1e-5)
+    """Config 3: training step via the Trainer (sequence loss, AdamW, AMP
+    bf16 autocast, channels-last, DP bucketed all-reduce when world>1)."""
+    from raft_amd.engine.trainer import Trainer, TrainConfig
     H = args.height if args.height != 436 else 368
     W = args.width if args.width != 1024 else 768
     b = args.batch if args.batch > 1 else 2
+    iters = args.iters if args.iters != 32 else 12
+    cfg = TrainConfig(num_steps=args.steps + args.warmup + 2, iters=iters,
+                      batch=b, height=H, width=W,
+                      amp=args.dtype == "bf16")
+    tr = Trainer(model, cfg, device=dev)
     x1 = torch.rand(b, 3, H, W, device=dev)
     x2 = torch.rand(b, 3, H, W, device=dev)
     gt = torch.randn(b, 2, H, W, device=dev)
 
     def step():
-        opt.zero_grad(set_to_none=True)
-        preds = model(x1, x2, iters=args.iters if args.iters != 32 else 12,
-                      test_mode=False)
-        loss = sequence_loss(preds, gt)
-        loss.backward()
-        if dist is not None:
-            model.finish_gradient_sync()
-        torch.nn.utils.clip_grad_norm_(model.parameters(), 1.0)
-        opt.step()
-        return loss
+        return tr.step(x1, x2, gt)
 
     return step, lambda: None
 

@@ -227,6 +227,16 @@ class FusedRaft:
         f2p = fmap2.permute(0, 2, 3, 1).contiguous()
         B, H8, W8, C = f1p.shape
 
+        # overlap: the context encoder (MIOpen) runs on a side stream
+        # concurrently with the correlation volume + pyramid build
+        side = getattr(self, "_side_stream", None)
+        if side is None:
+            side = torch.cuda.Stream()
+            self._side_stream = side
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            cnet = model.cnet(img1)
+
         vol = hip.corr_volume_nhwc(f1p, f2p, True)       # bf16 volume
         levels = [vol]
         for _ in range(cfg.corr_levels - 1):
@@ -236,7 +246,8 @@ class FusedRaft:
             else:
                 levels.append(hip.corr_pool2x_bf16(last))
 
-        cnet = model.cnet(img1)
+        torch.cuda.current_stream().wait_stream(side)
+        cnet.record_stream(torch.cuda.current_stream())
         net, inp = torch.split(cnet, [cfg.hidden_dim, cfg.context_dim], dim=1)
         net = torch.tanh(net).permute(0, 2, 3, 1).contiguous()
         inp = torch.relu(inp).permute(0, 2, 3, 1).contiguous()
