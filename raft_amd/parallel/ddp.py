@@ -23,16 +23,24 @@ import torch.distributed as dist
 import torch.nn as nn
 
 
-def init_distributed(backend: Optional[str] = None) -> int:
+def init_distributed(backend: Optional[str] = None,
+                     timeout_s: float = 600.0) -> int:
     """Initialize torch.distributed from torchrun env vars; returns rank.
-    No-op (returns 0) when WORLD_SIZE is absent/1."""
+    No-op (returns 0) when WORLD_SIZE is absent/1.
+
+    timeout_s bounds every collective: a dead rank aborts the job cleanly
+    (RCCL watchdog) instead of hanging — the failure-detection story for
+    single-node DP."""
+    import datetime
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world <= 1:
         return 0
     if not dist.is_initialized():
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        dist.init_process_group(
+            backend=backend,
+            timeout=datetime.timedelta(seconds=timeout_s))
     if torch.cuda.is_available():
         torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
     return dist.get_rank()
