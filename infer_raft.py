@@ -84,6 +84,7 @@ def mode_test(args, device):
     from raft_amd.engine.inference import InferenceEngine
     from raft_amd.data.imageio import write_png
     from raft_amd.utils.flow_viz import flow_to_color
+    from raft_amd.utils.flow_io import write_flo
 
     model = _build_model(args, device)
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
@@ -91,8 +92,18 @@ def mode_test(args, device):
                              use_graph=not args.no_graph)
     size = None if args.size == "native" else \
         tuple(int(v) for v in args.size.split("x"))
-    ds = PairDataflow([(args.im1, args.im2)], input_size=size,
-                      batch=args.batch)
+    if args.data and os.path.isdir(args.data):
+        # sequence mode: consecutive frame pairs from a directory
+        # (the reference parsed --data but never used it, infer_raft.py:54)
+        frames = sorted(
+            os.path.join(args.data, f) for f in os.listdir(args.data)
+            if f.lower().endswith(".png"))
+        pairs = list(zip(frames[:-1], frames[1:]))
+        if not pairs:
+            raise SystemExit(f"no consecutive PNG pairs in {args.data}")
+    else:
+        pairs = [(args.im1, args.im2)]
+    ds = PairDataflow(pairs, input_size=size, batch=args.batch)
     os.makedirs(args.out, exist_ok=True)
     variant = "raft-small" if args.small else "raft-things"
     for i, (im1, im2) in enumerate(ds):
@@ -104,8 +115,11 @@ def mode_test(args, device):
         print(i, tuple(flow.shape), f"{dt * 1e3:.1f} ms")
         flow_np = flow[0].float().permute(1, 2, 0).cpu().numpy()
         color = flow_to_color(flow_np, convert_to_bgr=True)
-        out_path = os.path.join(args.out, f"raft_flow_{variant}.png")
+        suffix = f"_{i:04d}" if len(pairs) > 1 else ""
+        out_path = os.path.join(args.out, f"raft_flow_{variant}{suffix}.png")
         write_png(out_path, color)
+        write_flo(os.path.join(args.out,
+                               f"raft_flow_{variant}{suffix}.flo"), flow_np)
         print(f"wrote {out_path}")
 
 

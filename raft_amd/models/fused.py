@@ -67,7 +67,7 @@ class _FC:
                  in1_off=0, in1_len=0):
         return hip.fconv_plain(in1, in2, self.wp, self.bias, self.kh,
                                self.kw, act, out, n_off, in1_off, in1_len,
-                               -1)
+                               -1, -1)
 
 
 class _GruDir:
@@ -118,7 +118,7 @@ class FusedBasicUpdate:
         net = self.gru2(hip, net, x_buf)
         # heads: one merged 3x3 conv, consumers read strided slices
         hbuf = hip.fconv_plain(net, None, self.heads_w, self.heads_b, 3, 3,
-                               ACT_RELU, None, 0, 0, 0, -1)
+                               ACT_RELU, None, 0, 0, 0, -1, -1)
         dflow = self.fh2(hip, hbuf, act=ACT_NONE, in1_off=0, in1_len=256)
         mask = self.m2(hip, hbuf, act=ACT_NONE, in1_off=256, in1_len=256)
         return net, mask, dflow

@@ -45,7 +45,8 @@ void launch_corr_pool2x_bf16(const void*, void*, int, int, int, int,
 void launch_fconv_nhwc_bf16(const void*, int, int, int, const void*, int,
                             const void*, const float*, void*, int, int, int,
                             int, int, int, int, int, int, int, const void*,
-                            const void*, void*, void*, int, hipStream_t);
+                            const void*, void*, void*, int, int,
+                            hipStream_t);
 void launch_fconv_smallk_nhwc_bf16(const void*, const void*, const float*,
                                    void*, int, int, int, int, int, int,
                                    int, int, hipStream_t);
@@ -285,7 +286,7 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
                        int64_t kh, int64_t kw, int64_t act,
                        c10::optional<at::Tensor> out_buf, int64_t n_off,
                        int64_t in1_off, int64_t in1_len,
-                       int64_t alltaps) {
+                       int64_t alltaps, int64_t mtiles) {
     CHECK_DEV(in1); CHECK_CONT(in1); CHECK_DEV(wp); CHECK_CONT(wp);
     TORCH_CHECK(in1.scalar_type() == at::kBFloat16, "fconv needs bf16");
     const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
@@ -326,7 +327,7 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
                            p2, C2, wp.data_ptr(), bptr, out.data_ptr(), B,
                            H, W, N, (int)n_off, cstride, (int)kh, (int)kw,
                            (int)act, 0, nullptr, nullptr, nullptr, nullptr,
-                           (int)alltaps, current_stream());
+                           (int)alltaps, (int)mtiles, current_stream());
     return out;
 }
 
@@ -343,7 +344,7 @@ std::vector<at::Tensor> fconv_gru_zr(at::Tensor h, at::Tensor x,
                            x.size(3), wp.data_ptr(), bias.data_ptr<float>(),
                            nullptr, B, H, W, N, 0, 0, (int)kh, (int)kw, 0,
                            1, h.data_ptr(), nullptr, z.data_ptr(),
-                           rh.data_ptr(), -1, current_stream());
+                           rh.data_ptr(), -1, -1, current_stream());
     return {z, rh};
 }
 
@@ -360,7 +361,7 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
                            x.size(3), wp.data_ptr(),
                            bias.data_ptr<float>(), out.data_ptr(), B, H, W,
                            hd, 0, hd, (int)kh, (int)kw, 0, 2, h.data_ptr(),
-                           z.data_ptr(), nullptr, nullptr, -1,
+                           z.data_ptr(), nullptr, nullptr, -1, -1,
                            current_stream());
     return out;
 }

@@ -68,7 +68,11 @@ RAFT_DEV float factivate(float v, int act) {
 // weight tile per K-step, so each barrier pair covers KH*KW MFMA groups
 // instead of KW — trades LDS (occupancy) for barrier amortization; only
 // meaningful for KH > 1 (the 5x1 GRU conv had 2 MFMA per barrier).
-template <int KH, int KW, int MI, int NJ, bool AT>
+// MT: m-tiles per block — each block computes MT consecutive 32*MI-wide
+// position tiles, re-using every staged weight tile MT times (the weight
+// slice is re-staged once per BLOCK per K-step; at batch-1 grids the
+// m-tile count is what multiplies that traffic).
+template <int KH, int KW, int MI, int NJ, bool AT, int MT>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
     int in1_stride, int in1_off,                      // strided slice of in1
@@ -86,7 +90,7 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int BN = 32 * NJ;          // block output channels
     constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
     constexpr int APAD = ((AW + 15) / 16) * 16;  // swizzle window rounding
-    constexpr int NSLAB = AT ? KH : 1;
+    constexpr int NSLAB = (AT ? KH : 1) * MT;
     constexpr int NBT = AT ? TAPS : KW;
     constexpr int ABYTES = APAD * FC_ROWB;
     constexpr int BBYTES = BN * FC_ROWB;
@@ -97,9 +101,9 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
-    const int tiles_per_row = (W + BM - 1) / BM;
+    const int tiles_per_row = (W + MT * BM - 1) / (MT * BM);
     const int y = blockIdx.y / tiles_per_row;
-    const int x0 = (blockIdx.y % tiles_per_row) * BM;
+    const int x0 = (blockIdx.y % tiles_per_row) * (MT * BM);
     const int n0 = blockIdx.x * BN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -110,11 +114,14 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const int ksteps = (Cin + FC_BK - 1) / FC_BK;
     const int nsteps = AT ? ksteps : KH * ksteps;
 
-    floatx4 acc[MI][NJ];
+    floatx4 acc[MT][MI][NJ];
 #pragma unroll
-    for (int i = 0; i < MI; ++i)
+    for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
-        for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int i = 0; i < MI; ++i)
+#pragma unroll
+            for (int j = 0; j < NJ; ++j)
+                acc[mt][i][j] = {0.f, 0.f, 0.f, 0.f};
 
     char* const sAbase = smem;
     char* const sBbase = smem + NSLAB * ABYTES;
@@ -124,16 +131,18 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     auto stage = [&](int s) {
         const int ty0 = AT ? 0 : s / ksteps;
         const int k0 = (AT ? s : (s - ty0 * ksteps)) * FC_BK;
-        // ---- A slabs: NSLAB x AW rows x FC_BK ch (16 B per thread pass)
+        // ---- A slabs: NSLAB (= rows x m-tiles) x AW x FC_BK ch
         for (int e = tid; e < NSLAB * AW * (FC_BK / 8); e += 256) {
             const int sl = e / (AW * (FC_BK / 8));
             const int rem0 = e % (AW * (FC_BK / 8));
             const int ar = rem0 / (FC_BK / 8);
             const int c8 = (rem0 % (FC_BK / 8)) * 8;
             char* sA = sAbase + sl * ABYTES;
-            const int row = y + (ty0 + sl) - KH / 2;
+            const int mt = sl % MT;              // m-tile index
+            const int rsl = sl / MT;             // kernel-row slab index
+            const int row = y + (ty0 + rsl) - KH / 2;
             const bool row_ok = (row >= 0 && row < H);
-            const int x = x0 + ar - KW / 2;
+            const int x = x0 + mt * BM + ar - KW / 2;
             uint4v v = {0, 0, 0, 0};
             if (row_ok && x >= 0 && x < W) {
                 const int k = k0 + c8;
@@ -199,26 +208,30 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
         __syncthreads();
         const unsigned cb = (lane >> 4) * 16;
 #pragma unroll
-        for (int sl = 0; sl < NSLAB; ++sl) {
-            const char* sA = sAbase + sl * ABYTES;
+        for (int rsl = 0; rsl < (AT ? KH : 1); ++rsl) {
 #pragma unroll
-            for (int tx = 0; tx < KW; ++tx) {
-                short8 af[MI], bf[NJ];
+            for (int mt = 0; mt < MT; ++mt) {
+                const char* sA = sAbase + (rsl * MT + mt) * ABYTES;
 #pragma unroll
-                for (int i = 0; i < MI; ++i)
-                    af[i] = *(const short8*)(
-                        sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+                for (int tx = 0; tx < KW; ++tx) {
+                    short8 af[MI], bf[NJ];
 #pragma unroll
-                for (int j = 0; j < NJ; ++j)
-                    bf[j] = *(const short8*)(
-                        sBbase + (sl * KW + tx) * BBYTES +
-                        fswz(wn + j * 16 + (lane & 15), cb));
-#pragma unroll
-                for (int i = 0; i < MI; ++i)
+                    for (int i = 0; i < MI; ++i)
+                        af[i] = *(const short8*)(
+                            sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
 #pragma unroll
                     for (int j = 0; j < NJ; ++j)
-                        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                            af[i], bf[j], acc[i][j], 0, 0, 0);
+                        bf[j] = *(const short8*)(
+                            sBbase + (rsl * KW + tx) * BBYTES +
+                            fswz(wn + j * 16 + (lane & 15), cb));
+#pragma unroll
+                    for (int i = 0; i < MI; ++i)
+#pragma unroll
+                        for (int j = 0; j < NJ; ++j)
+                            acc[mt][i][j] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    af[i], bf[j], acc[mt][i][j], 0, 0, 0);
+                }
             }
         }
         __syncthreads();
@@ -227,6 +240,8 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     // ------------------------------------------------------------- epilogue
     const int hd = (mode == EP_GRU_ZR) ? N / 2 : N;
 #pragma unroll
+    for (int mt = 0; mt < MT; ++mt)
+#pragma unroll
     for (int i = 0; i < MI; ++i)
 #pragma unroll
         for (int j = 0; j < NJ; ++j)
@@ -234,10 +249,10 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             for (int r = 0; r < 4; ++r) {
                 const int m = wm + i * 16 + (lane >> 4) * 4 + r;
                 const int n = n0 + wn + j * 16 + (lane & 15);
-                const int x = x0 + m;
+                const int x = x0 + mt * BM + m;
                 if (x >= W || n >= N) continue;
                 const long long p = ((long long)b * H + y) * W + x;
-                float v = acc[i][j][r];
+                float v = acc[mt][i][j][r];
                 if (bias) v += bias[n];
                 if (mode == EP_PLAIN) {
                     out[p * out_cstride + n_off + n] =
@@ -328,7 +343,7 @@ extern "C" void launch_fconv_nhwc_bf16(
     int C2, const void* wp, const float* bias, void* out, int B, int H,
     int W, int N, int n_off, int out_cstride, int kh, int kw, int act,
     int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
-    void* rh_out, int alltaps, hipStream_t s) {
+    void* rh_out, int alltaps, int mtiles, hipStream_t s) {
     dim3 blk(256);
     const long long big_blocks0 =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
@@ -355,21 +370,27 @@ extern "C" void launch_fconv_nhwc_bf16(
             return;
         }
     }
-#define FC_LAUNCH(KH, KW, MI, NJ, AT, BMv, BNv)                              \
+    // mtiles: -1 auto (2 for the small tile — halves the per-block weight
+    // re-staging with grid still >= ~440 blocks at 55x128), else 1/2/4.
+    const int mt = (mtiles < 0) ? 2 : mtiles;
+#define FC_LAUNCH(KH, KW, MI, NJ, AT, MTv, BMv, BNv)                         \
     {                                                                        \
-        dim3 grid(cdiv(N, BNv), H * cdiv(W, BMv), B);                        \
-        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, MI, NJ, AT>), grid,    \
-                           blk, 0, s, FCONV_ARGS);                           \
+        dim3 grid(cdiv(N, BNv), H * cdiv(W, (BMv) * (MTv)), B);              \
+        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, MI, NJ, AT, MTv>),     \
+                           grid, blk, 0, s, FCONV_ARGS);                     \
         return;                                                              \
     }
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
+        if (big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)                  \
         if (KH > 1 && at) {                                                  \
-            if (big) FC_LAUNCH(KH, KW, 2, 4, true, 64, 128)                  \
-            else FC_LAUNCH(KH, KW, 1, 2, true, 32, 64)                       \
+            if (mt >= 4) FC_LAUNCH(KH, KW, 1, 2, true, 4, 32, 64)            \
+            if (mt == 2) FC_LAUNCH(KH, KW, 1, 2, true, 2, 32, 64)            \
+            FC_LAUNCH(KH, KW, 1, 2, true, 1, 32, 64)                         \
         }                                                                    \
-        if (big) FC_LAUNCH(KH, KW, 2, 4, false, 64, 128)                     \
-        else FC_LAUNCH(KH, KW, 1, 2, false, 32, 64)                          \
+        if (mt >= 4) FC_LAUNCH(KH, KW, 1, 2, false, 4, 32, 64)               \
+        if (mt == 2) FC_LAUNCH(KH, KW, 1, 2, false, 2, 32, 64)               \
+        FC_LAUNCH(KH, KW, 1, 2, false, 1, 32, 64)                            \
     }
     FC_CASE(1, 1)
     FC_CASE(3, 3)

@@ -112,3 +112,19 @@ def test_cli_export_mode(tmp_path):
     assert os.path.exists(tmp_path / "raft-small.npz")
     # exported npz reloads
     infer_raft.main(["--mode", "flops", "--small"])
+
+
+def test_cli_test_mode_sequence_dir(tmp_path):
+    """--data directory: consecutive-pair sequence processing."""
+    import infer_raft
+    img = (np.random.rand(40, 56, 3) * 255).astype(np.uint8)
+    seq = tmp_path / "seq"
+    seq.mkdir()
+    from raft_amd.data.imageio import write_png as wp
+    for i in range(3):
+        wp(str(seq / f"frame_{i:02d}.png"), img)
+    out = tmp_path / "out"
+    infer_raft.main(["--mode", "test", "--small", "--data", str(seq),
+                     "--out", str(out), "--size", "40x56", "--iters", "2"])
+    assert (out / "raft_flow_raft-small_0000.png").exists()
+    assert (out / "raft_flow_raft-small_0001.flo").exists()

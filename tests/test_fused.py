@@ -40,7 +40,7 @@ def test_fconv_matches_conv2d(dev, kh, kw):
     w = torch.randn(N, Cin, kh, kw, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, kh, kw,
-                             1, None, 0, 0, 0, -1)
+                             1, None, 0, 0, 0, -1, -1)
     # reference: fp32 conv on the bf16-rounded inputs/weights
     xr = x.float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias,
@@ -59,7 +59,7 @@ def test_fconv_two_inputs_and_slice_output(dev):
     bias = torch.zeros(N, device=dev)
     buf = torch.zeros(B, H, W, 80, device=dev, dtype=torch.bfloat16)
     out = _hip().fconv_plain(a.contiguous(), b.contiguous(), _pack(w), bias,
-                             3, 3, 0, buf, 16, 0, 0, -1)
+                             3, 3, 0, buf, 16, 0, 0, -1, -1)
     assert out.data_ptr() == buf.data_ptr()
     xr = torch.cat([a, b], dim=-1).float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias, padding=1)
@@ -77,7 +77,7 @@ def test_fconv_small_cin_seam(dev):
     w = torch.randn(N, 12, 3, 3, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, 3, 3,
-                             1, None, 0, 0, 0, -1)
+                             1, None, 0, 0, 0, -1, -1)
     ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=1))
     ref = ref.permute(0, 2, 3, 1)
@@ -213,7 +213,7 @@ def test_fconv_strided_input_slice(dev):
     w = torch.randn(24, 32, 3, 3, device=dev) * 0.1
     bias = torch.zeros(24, device=dev)
     out = _hip().fconv_plain(buf.contiguous(), None, _pack(w), bias, 3, 3,
-                             0, None, 0, 40, 32, -1)
+                             0, None, 0, 40, 32, -1, -1)
     ref = F.conv2d(buf[..., 40:72].float().permute(0, 3, 1, 2),
                    w.to(torch.bfloat16).float(), bias, padding=1)
     ref = ref.permute(0, 2, 3, 1)

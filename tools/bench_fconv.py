@@ -48,7 +48,9 @@ def main():
         ("m2 1x1", 256, 576, 1, 1),
     ]
     total = {0: 0.0, 1: 0.0}
-    print(f"{'conv':<14} {'GF':>6} {'AT=0 us':>9} {'AT=1 us':>9} {'TF@best':>8}")
+    print(f"{'conv':<14} {'GF':>6} "
+          f"{'a0m1':>7} {'a0m2':>7} {'a0m4':>7} "
+          f"{'a1m1':>7} {'a1m2':>7} {'a1m4':>7} {'TF@bst':>7}")
     for name, cin, n, kh, kw in configs:
         x = torch.randn(B, H, W, cin, device=dev).to(torch.bfloat16)
         wp = torch.randn(kh * kw, n, cin, device=dev).to(torch.bfloat16) * 0.1
@@ -56,12 +58,14 @@ def main():
         gf = 2.0 * B * H * W * n * cin * kh * kw / 1e9
         res = {}
         for at in (0, 1):
-            res[at] = bench(lambda: hip.fconv_plain(
-                x, None, wp, bias, kh, kw, 1, None, 0, 0, 0, at))
-            total[at] += res[at]
+            for mt in (1, 2, 4):
+                res[(at, mt)] = bench(lambda: hip.fconv_plain(
+                    x, None, wp, bias, kh, kw, 1, None, 0, 0, 0, at, mt))
+            total[at] += res[(at, 2)]
         best = min(res.values())
-        print(f"{name:<14} {gf:>6.2f} {res[0]:>9.1f} {res[1]:>9.1f} "
-              f"{gf / best * 1e6 / 1e3:>7.0f}")
+        cells = " ".join(f"{res[(a, m)]:>7.1f}" for a in (0, 1)
+                         for m in (1, 2, 4))
+        print(f"{name:<14} {gf:>6.2f} {cells} {gf/best*1e6/1e3:>7.0f}")
     print(f"{'TOTAL':<14} {'':>6} {total[0]:>9.1f} {total[1]:>9.1f}")
 
     # GRU fused pair timing (zr + q as used by the model)
