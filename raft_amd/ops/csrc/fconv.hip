@@ -370,9 +370,11 @@ extern "C" void launch_fconv_nhwc_bf16(
             return;
         }
     }
-    // mtiles: -1 auto (2 for the small tile — halves the per-block weight
-    // re-staging with grid still >= ~440 blocks at 55x128), else 1/2/4.
-    const int mt = (mtiles < 0) ? 2 : mtiles;
+    // mtiles: -1 auto = 1. MT>1 was hypothesized to win by amortizing
+    // per-block weight staging but MEASURED worse nearly everywhere
+    // (tools/bench_fconv.py round14: grid saturation dominates at these
+    // batch-1 sizes) — kept selectable for larger-batch shapes.
+    const int mt = (mtiles < 0) ? 1 : mtiles;
 #define FC_LAUNCH(KH, KW, MI, NJ, AT, MTv, BMv, BNv)                         \
     {                                                                        \
         dim3 grid(cdiv(N, BNv), H * cdiv(W, (BMv) * (MTv)), B);              \
