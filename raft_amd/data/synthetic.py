@@ -7,7 +7,6 @@ flow_gt) triplets with exact ground truth.
 """
 from __future__ import annotations
 
-import math
 from typing import Tuple
 
 import torch

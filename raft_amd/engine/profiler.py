@@ -2,7 +2,6 @@
 crashing flops mode, infer_raft.py:80-95) and simple step timers."""
 from __future__ import annotations
 
-import math
 import time
 from typing import Optional
 

@@ -21,7 +21,7 @@ Weight packing: conv weights [N, Cin, kh, kw] -> [kh*kw, N, Cin_pad] bf16
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch
 

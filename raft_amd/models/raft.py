@@ -18,7 +18,7 @@ ops.torch_ref.upflow8).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 import torch

@@ -1,6 +1,4 @@
 """npz checkpoint contract tests (SURVEY.md §5.4 key schema)."""
-import os
-
 import numpy as np
 import torch
 

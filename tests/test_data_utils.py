@@ -2,7 +2,6 @@
 import os
 
 import numpy as np
-import pytest
 import torch
 
 from raft_amd.data.imageio import read_png, write_png

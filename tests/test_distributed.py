@@ -1,7 +1,6 @@
 """Multi-process DP tests on gloo (world_size 2, CPU) — the distributed
 path must be correct by construction before it ever touches RCCL/xGMI."""
 import os
-import sys
 
 import pytest
 import torch

@@ -1,5 +1,4 @@
 """Model-level shape / parameter / mode tests (CPU, golden path)."""
-import pytest
 import torch
 
 from raft_amd import RAFT, RaftConfig

@@ -8,7 +8,6 @@ convex upsample (networks/RAFT.py:119-134).
 """
 import math
 
-import numpy as np
 import torch
 
 from raft_amd.ops import torch_ref as R

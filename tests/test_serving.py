@@ -3,7 +3,6 @@ import io
 
 import numpy as np
 import pytest
-import torch
 
 pytest.importorskip("fastapi")
 from starlette.testclient import TestClient
