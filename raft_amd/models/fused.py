@@ -104,24 +104,27 @@ class FusedBasicUpdate:
         self.m2 = _FC(ub.mask[2], scale=0.25)        # fold the 0.25 scale
         self.ctx_dim = ctx_dim
 
-    def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
-        # motion encoder (model_utils.py:110-119), outputs into x_buf slice
-        cor = self.c2(hip, self.c1(hip, corr_pad))
-        flo1 = hip.fconv_smallk(flow_bf, self.f1.wp, self.f1.bias,
-                                self.f1.kh, self.f1.kw, ACT_RELU)
-        flo = self.f2(hip, flo1)
+    def __call__(self, hip, net, x_buf, corr_pad, coords1):
+        # motion encoder (model_utils.py:110-119). The flow channels of
+        # x_buf were already written by the lookup kernel (fused flow out).
         ctx = self.ctx_dim
+        cor = self.c2(hip, self.c1(hip, corr_pad))
+        flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                self.f1.kh, self.f1.kw, ACT_RELU,
+                                ctx + 126, 2)
+        flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
-        x_buf[..., ctx + 126:ctx + 128] = flow_bf
         # SepConvGRU (model_utils.py:138-156)
         net = self.gru1(hip, net, x_buf)
         net = self.gru2(hip, net, x_buf)
-        # heads: one merged 3x3 conv, consumers read strided slices
+        # heads: one merged 3x3 conv; mask from a strided slice; the
+        # delta-flow final conv also applies coords1 += dflow in-kernel
         hbuf = hip.fconv_plain(net, None, self.heads_w, self.heads_b, 3, 3,
                                ACT_RELU, None, 0, 0, 0, -1, -1)
-        dflow = self.fh2(hip, hbuf, act=ACT_NONE, in1_off=0, in1_len=256)
+        coords_new = hip.fconv_dflow_coords(hbuf, self.fh2.wp,
+                                            self.fh2.bias, coords1, 3, 3)
         mask = self.m2(hip, hbuf, act=ACT_NONE, in1_off=256, in1_len=256)
-        return net, mask, dflow
+        return net, mask, coords_new
 
 
 class FusedSmallUpdate:
@@ -139,17 +142,19 @@ class FusedSmallUpdate:
         self.ctx_dim = ctx_dim
         # x = [inp(ctx) | motion(80) | flow(2)]; motion encoder out = 80
 
-    def __call__(self, hip, net, x_buf, corr_pad, flow_bf):
-        cor = self.c1(hip, corr_pad)
-        flo1 = hip.fconv_smallk(flow_bf, self.f1.wp, self.f1.bias,
-                                self.f1.kh, self.f1.kw, ACT_RELU)
-        flo = self.f2(hip, flo1)
+    def __call__(self, hip, net, x_buf, corr_pad, coords1):
         ctx = self.ctx_dim
+        cor = self.c1(hip, corr_pad)
+        flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                self.f1.kh, self.f1.kw, ACT_RELU,
+                                ctx + 80, 2)
+        flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
-        x_buf[..., ctx + 80:ctx + 82] = flow_bf
         net = self.gru(hip, net, x_buf)
-        dflow = self.fh2(hip, self.fh1(hip, net), act=ACT_NONE)
-        return net, None, dflow
+        h1 = self.fh1(hip, net)
+        coords_new = hip.fconv_dflow_coords(h1, self.fh2.wp, self.fh2.bias,
+                                            coords1, 3, 3)
+        return net, None, coords_new
 
 
 class FusedRaft:
@@ -190,17 +195,16 @@ class FusedRaft:
         B, H8, W8, _ = coords0.shape
         corr_buf = torch.zeros(B, H8, W8, self.corr_cpad,
                                device=net.device, dtype=torch.bfloat16)
-        flow_buf = torch.empty(B, H8, W8, 2, device=net.device,
-                               dtype=torch.bfloat16)
+        flow_off = self.x_dim - 2      # flow channels of the GRU input
         mask = None
         for _ in range(iters):
-            # lookup also emits flow = coords1 - identity grid (fused)
+            # lookup writes the taps AND flow = coords1 - grid directly
+            # into the GRU input buffer's flow slice (zero glue kernels)
             corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
                                             cfg.corr_radius, self.corr_cpad,
-                                            True, corr_buf, flow_buf)
-            flow_bf = flow_buf
-            net, mask, dflow = self.update(hip, net, x_buf, corr_pad, flow_bf)
-            coords1 = coords1 + dflow.float()
+                                            True, corr_buf, x_buf, flow_off)
+            net, mask, coords1 = self.update(hip, net, x_buf, corr_pad,
+                                             coords1)
 
         flow = coords1 - coords0                         # [B,H,W,2] fp32
         if cfg.small:

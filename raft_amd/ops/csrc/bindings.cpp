@@ -36,7 +36,8 @@ void launch_corr_volume_nhwc_bf16(const void*, const void*, void*, bool,
                                   int, int, int, int, float, hipStream_t);
 void launch_corr_lookup_nhwc(const void* const*, const int*, const int*,
                              bool, const float*, void*, bool, void*, int,
-                             int, int, int, int, int, hipStream_t);
+                             int, int, int, int, int, int, int,
+                             hipStream_t);
 void launch_corr_lookup_nhwc_bwd(float* const*, const int*, const int*,
                                  const float*, const float*, int, int, int,
                                  int, int, hipStream_t);
@@ -47,9 +48,12 @@ void launch_fconv_nhwc_bf16(const void*, int, int, int, const void*, int,
                             int, int, int, int, int, int, int, const void*,
                             const void*, void*, void*, int, int,
                             hipStream_t);
-void launch_fconv_smallk_nhwc_bf16(const void*, const void*, const float*,
-                                   void*, int, int, int, int, int, int,
-                                   int, int, hipStream_t);
+void launch_fconv_smallk_nhwc_bf16(const void*, int, int, const void*,
+                                   const float*, void*, int, int, int, int,
+                                   int, int, int, int, hipStream_t);
+void launch_fconv_dflow_coords(const void*, int, int, int, const void*,
+                               const float*, const float*, float*, int, int,
+                               int, int, int, hipStream_t);
 }
 
 namespace {
@@ -236,7 +240,8 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
                             at::Tensor coords, int64_t radius,
                             int64_t c_stride, bool out_bf16,
                             c10::optional<at::Tensor> out_buf,
-                            c10::optional<at::Tensor> flow_buf) {
+                            c10::optional<at::Tensor> flow_buf,
+                            int64_t flow_off) {
     // coords: [B, H, W, 2] fp32; returns physical NHWC [B, H, W, c_stride]
     // with channels [L*KK..c_stride) zero-filled (pad for fconv Cin%8).
     CHECK_DEV(coords); CHECK_CONT(coords);
@@ -269,15 +274,18 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
             : at::zeros({B, H, W, (int64_t)c_stride}, opts);
     }
     void* fptr = nullptr;
+    int fstride = 2;
     if (flow_buf.has_value()) {
         TORCH_CHECK(flow_buf->is_contiguous() &&
                     flow_buf->scalar_type() == at::kBFloat16);
         fptr = flow_buf->data_ptr();
+        fstride = (int)flow_buf->size(3);
     }
     launch_corr_lookup_nhwc(ptrs, hs, ws, vol_bf16,
                             coords.data_ptr<float>(), out.data_ptr(),
-                            out_bf16, fptr, B, H, W, L, (int)radius,
-                            (int)c_stride, current_stream());
+                            out_bf16, fptr, fstride, (int)flow_off, B, H, W,
+                            L, (int)radius, (int)c_stride,
+                            current_stream());
     return out;
 }
 
@@ -368,16 +376,36 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
 
 at::Tensor fconv_smallk(at::Tensor in1, at::Tensor wp,
                         c10::optional<at::Tensor> bias, int64_t kh,
-                        int64_t kw, int64_t act) {
+                        int64_t kw, int64_t act, int64_t in1_off,
+                        int64_t in1_len) {
     CHECK_DEV(in1); CHECK_CONT(in1); CHECK_CONT(wp);
     const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
-    const int C = in1.size(3);
+    const int stride = in1.size(3);
+    const int C = in1_len > 0 ? (int)in1_len : stride;
     const int N = wp.size(1);
     const float* bptr = bias.has_value() ? bias->data_ptr<float>() : nullptr;
     auto out = at::empty({B, H, W, N}, in1.options());
-    launch_fconv_smallk_nhwc_bf16(in1.data_ptr(), wp.data_ptr(), bptr,
-                                  out.data_ptr(), B, H, W, C, N, (int)kh,
-                                  (int)kw, (int)act, current_stream());
+    launch_fconv_smallk_nhwc_bf16(in1.data_ptr(), stride, (int)in1_off,
+                                  wp.data_ptr(), bptr, out.data_ptr(), B, H,
+                                  W, C, N, (int)kh, (int)kw, (int)act,
+                                  current_stream());
+    return out;
+}
+
+at::Tensor fconv_dflow_coords(at::Tensor in1, at::Tensor wp, at::Tensor bias,
+                              at::Tensor coords, int64_t kh, int64_t kw) {
+    // delta-flow 3x3 -> 2 head with the coords update fused:
+    // returns coords_new = coords + conv(in1)
+    CHECK_DEV(in1); CHECK_CONT(in1); CHECK_CONT(wp); CHECK_CONT(coords);
+    const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
+    const int Cin = wp.size(2);
+    TORCH_CHECK(wp.size(1) == 2 && coords.scalar_type() == at::kFloat);
+    auto out = at::empty_like(coords);
+    launch_fconv_dflow_coords(in1.data_ptr(), Cin, in1.size(3), 0,
+                              wp.data_ptr(), bias.data_ptr<float>(),
+                              coords.data_ptr<float>(),
+                              out.data_ptr<float>(), B, H, W, (int)kh,
+                              (int)kw, current_stream());
     return out;
 }
 
@@ -402,5 +430,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused NHWC bf16 conv (+bias +activation, slice output)");
     m.def("fconv_gru_zr", &fconv_gru_zr, "GRU z/r gate conv pair");
     m.def("fconv_smallk", &fconv_smallk, "direct NHWC conv for tiny Cin");
+    m.def("fconv_dflow_coords", &fconv_dflow_coords,
+          "flow-head final conv fused with the coords update");
     m.def("fconv_gru_q", &fconv_gru_q, "GRU candidate conv + state update");
 }

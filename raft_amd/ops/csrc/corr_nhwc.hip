@@ -146,7 +146,8 @@ template <typename T, typename OT>
 __global__ void corr_lookup_nhwc_k(
     LevelsT lv, const float* __restrict__ coords,  // [B, H, W, 2]
     OT* __restrict__ out,                          // [B, H, W, Cs]
-    __hip_bfloat16* __restrict__ flow_out,         // [B, H, W, 2] or null
+    __hip_bfloat16* __restrict__ flow_out,         // strided slice or null
+    int flow_stride, int flow_off,
     int H, int W, int num_levels, int radius, int Cs, long long total) {
     const int K = 2 * radius + 1;
     const int KK = K * K;
@@ -161,10 +162,12 @@ __global__ void corr_lookup_nhwc_k(
 
         if (flow_out && c < 2) {
             // fused flow = coords - identity grid (coords0 is the pixel
-            // grid by construction, RAFT.py:111-117)
+            // grid by construction, RAFT.py:111-117); written into a
+            // channel slice of the GRU input buffer directly
             const float base = (c == 0) ? (float)(q % W)
                                         : (float)((q / W) % H);
-            flow_out[q * 2 + c] = (__hip_bfloat16)(coords[q * 2 + c] - base);
+            flow_out[q * flow_stride + flow_off + c] =
+                (__hip_bfloat16)(coords[q * 2 + c] - base);
         }
 
         const float inv = 1.0f / (float)(1 << lvl);
@@ -185,8 +188,8 @@ __global__ void corr_lookup_nhwc_k(
 extern "C" void launch_corr_lookup_nhwc(
     const void* const* level_ptrs, const int* level_h, const int* level_w,
     bool vol_bf16, const float* coords, void* out, bool out_bf16,
-    void* flow_out, int B, int H, int W, int num_levels, int radius, int Cs,
-    hipStream_t s) {
+    void* flow_out, int flow_stride, int flow_off, int B, int H, int W,
+    int num_levels, int radius, int Cs, hipStream_t s) {
     LevelsT lv{};
     for (int i = 0; i < num_levels; ++i) {
         lv.ptr[i] = level_ptrs[i];
@@ -201,8 +204,8 @@ extern "C" void launch_corr_lookup_nhwc(
         out_bf16 == OTC) {                                                  \
         hipLaunchKernelGGL((corr_lookup_nhwc_k<T, OT>), dim3(blocks),       \
                            dim3(256), 0, s, lv, coords, (OT*)out,           \
-                           (__hip_bfloat16*)flow_out, H, W,                 \
-                           num_levels, radius, Cs, total);                  \
+                           (__hip_bfloat16*)flow_out, flow_stride,          \
+                           flow_off, H, W, num_levels, radius, Cs, total);  \
         return;                                                             \
     }
     LKCASE(float, float, false)

@@ -286,7 +286,9 @@ __global__ __launch_bounds__(256) void fconv_tinyn_k(
     int Cin, int in_stride, int in_off,
     const __hip_bfloat16* __restrict__ wp,    // [taps][N][Cin]
     const float* __restrict__ bias,
-    __hip_bfloat16* __restrict__ out,         // [B,H,W,N]
+    __hip_bfloat16* __restrict__ out,         // [B,H,W,N] (null if coords)
+    const float* __restrict__ coords_in,      // fused coords1 += dflow
+    float* __restrict__ coords_out,           // (NN == 2 only)
     int H, int W, int kh, int kw, int act, long long ncells) {
     const int lane = threadIdx.x & 63;
     const long long cell = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
@@ -324,7 +326,11 @@ __global__ __launch_bounds__(256) void fconv_tinyn_k(
 #pragma unroll
         for (int n = 0; n < NN; ++n) {
             float v = acc[n] + (bias ? bias[n] : 0.0f);
-            out[cell * NN + n] = (__hip_bfloat16)factivate(v, act);
+            v = factivate(v, act);
+            if (coords_out)   // fused coords1 = coords1 + delta_flow
+                coords_out[cell * NN + n] = coords_in[cell * NN + n] + v;
+            else
+                out[cell * NN + n] = (__hip_bfloat16)v;
         }
     }
 }
@@ -365,8 +371,8 @@ extern "C" void launch_fconv_nhwc_bf16(
             hipLaunchKernelGGL(fconv_tinyn_k<2>, tg, blk, 0, s,
                                (const __hip_bfloat16*)in1, C1, in1_stride,
                                in1_off, (const __hip_bfloat16*)wp, bias,
-                               (__hip_bfloat16*)out, H, W, kh, kw, act,
-                               ncells);
+                               (__hip_bfloat16*)out, nullptr, nullptr, H, W,
+                               kh, kw, act, ncells);
             return;
         }
     }
@@ -412,7 +418,8 @@ extern "C" void launch_fconv_nhwc_bf16(
 #define SK_MAXIN (8 * 49 * 4)        // PPB x taps x C cap
 
 extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
-    const __hip_bfloat16* __restrict__ in,    // [B, H, W, C]
+    const __hip_bfloat16* __restrict__ in,    // [B, H, W, *] slice
+    int in_stride, int in_off,
     const __hip_bfloat16* __restrict__ wp,    // [kh*kw][N][C]
     const float* __restrict__ bias,
     __hip_bfloat16* __restrict__ out,         // [B, H, W, N]
@@ -443,7 +450,8 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
                 const int yy = y + t / kw - kh / 2;
                 const int xx = x + t % kw - kw / 2;
                 if (yy >= 0 && yy < H && xx >= 0 && xx < W)
-                    v = in[(((long long)b * H + yy) * W + xx) * C + c];
+                    v = in[(((long long)b * H + yy) * W + xx) * in_stride
+                           + in_off + c];
             }
             sin[e] = v;
         }
@@ -466,7 +474,7 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
 
 // grid-stride naive fallback for shapes outside the LDS caps
 extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
-    const __hip_bfloat16* __restrict__ in,
+    const __hip_bfloat16* __restrict__ in, int in_stride, int in_off,
     const __hip_bfloat16* __restrict__ wp,
     const float* __restrict__ bias,
     __hip_bfloat16* __restrict__ out,
@@ -487,7 +495,8 @@ extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
                 const int xx = x + tx - kw / 2;
                 if (xx < 0 || xx >= W) continue;
                 const __hip_bfloat16* src =
-                    in + (((long long)b * H + yy) * W + xx) * C;
+                    in + (((long long)b * H + yy) * W + xx) * in_stride
+                    + in_off;
                 const __hip_bfloat16* wr =
                     wp + ((size_t)(ty * kw + tx) * N + n) * C;
                 for (int c = 0; c < C; ++c)
@@ -499,17 +508,22 @@ extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
 }
 
 extern "C" void launch_fconv_smallk_nhwc_bf16(
-    const void* in, const void* wp, const float* bias, void* out, int B,
-    int H, int W, int C, int N, int kh, int kw, int act, hipStream_t s) {
+    const void* in, int in_stride, int in_off, const void* wp,
+    const float* bias, void* out, int B, int H, int W, int C, int N, int kh,
+    int kw, int act, hipStream_t s) {
     const long long ncells = (long long)B * H * W;
     const int taps = kh * kw;
     if (N <= 128 && 256 % N == 0 && taps * N * C <= SK_MAXW &&
         (256 / N) * taps * C <= SK_MAXIN) {
         const int PPB = 256 / N;
         const long long groups = (ncells + PPB - 1) / PPB;
-        int blocks = (int)min(groups, (long long)4096);
+        // cap the grid well below the group count: every block stages the
+        // full weight set once (25 KB for convf1) — 3520 one-group blocks
+        // cost 88 MB of weight reads (measured 43.5 us); ~640 blocks
+        // amortize it ~6x while still filling 256 CUs.
+        int blocks = (int)min(groups, (long long)640);
         hipLaunchKernelGGL(fconv_smallk_lds_k, dim3(blocks), dim3(256), 0,
-                           s, (const __hip_bfloat16*)in,
+                           s, (const __hip_bfloat16*)in, in_stride, in_off,
                            (const __hip_bfloat16*)wp, bias,
                            (__hip_bfloat16*)out, H, W, C, N, kh, kw, act,
                            ncells);
@@ -518,7 +532,23 @@ extern "C" void launch_fconv_smallk_nhwc_bf16(
     const long long total = ncells * N;
     int blocks = (int)min((total + 255) / 256, (long long)4096);
     hipLaunchKernelGGL(fconv_smallk_nhwc_bf16_k, dim3(blocks), dim3(256), 0,
-                       s, (const __hip_bfloat16*)in,
+                       s, (const __hip_bfloat16*)in, in_stride, in_off,
                        (const __hip_bfloat16*)wp, bias,
                        (__hip_bfloat16*)out, H, W, C, N, kh, kw, act, total);
+}
+
+
+// delta-flow head with the coords update fused: coords_out = coords_in +
+// conv(in) — removes the per-iteration cast+add (aten::copy_/add_ glue,
+// ~30 us/iter measured in the chrome trace).
+extern "C" void launch_fconv_dflow_coords(
+    const void* in, int Cin, int in_stride, int in_off, const void* wp,
+    const float* bias, const float* coords_in, float* coords_out, int B,
+    int H, int W, int kh, int kw, hipStream_t s) {
+    const long long ncells = (long long)B * H * W;
+    dim3 tg((unsigned)((ncells + 3) / 4));
+    hipLaunchKernelGGL(fconv_tinyn_k<2>, tg, dim3(256), 0, s,
+                       (const __hip_bfloat16*)in, Cin, in_stride, in_off,
+                       (const __hip_bfloat16*)wp, bias, nullptr, coords_in,
+                       coords_out, H, W, kh, kw, 0, ncells);
 }

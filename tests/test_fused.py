@@ -126,7 +126,7 @@ def test_corr_lookup_nhwc_matches_ref(dev):
     coords = torch.rand(B, H, W, 2, device=dev) * 14.0 - 1.0
     C = 4 * 81
     cpad = 328
-    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False, None, None)
+    out = _hip().corr_lookup_nhwc(list(pyr), coords, r, cpad, False, None, None, 0)
     assert out.shape == (B, H, W, cpad)
     ref = R.corr_lookup(pyr, coords, r).permute(0, 2, 3, 1)
     assert torch.allclose(out[..., :C], ref, atol=1e-4, rtol=1e-4)
@@ -180,7 +180,7 @@ def test_fconv_smallk_matches_conv2d(dev):
     x = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
     w = torch.randn(N, C, 7, 7, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
-    out = _hip().fconv_smallk(x.contiguous(), _pack(w), bias, 7, 7, 1)
+    out = _hip().fconv_smallk(x.contiguous(), _pack(w), bias, 7, 7, 1, 0, 0)
     ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=3))
     ref = ref.permute(0, 2, 3, 1)
@@ -228,10 +228,38 @@ def test_lookup_fused_flow_output(dev):
     coords = torch.rand(B, H, W, 2, device=dev) * 8.0
     flow_buf = torch.empty(B, H, W, 2, device=dev, dtype=torch.bfloat16)
     C = 3 * 25
-    _hip().corr_lookup_nhwc(list(pyr), coords, r, C, False, None, flow_buf)
+    _hip().corr_lookup_nhwc(list(pyr), coords, r, C, False, None, flow_buf, 0)
     ys, xs = torch.meshgrid(torch.arange(H, device=dev, dtype=torch.float32),
                             torch.arange(W, device=dev, dtype=torch.float32),
                             indexing="ij")
     grid = torch.stack([xs, ys], dim=-1)[None]
     ref = (coords - grid).to(torch.bfloat16)
     assert torch.equal(flow_buf, ref)
+
+
+def test_fconv_smallk_slice_input(dev):
+    """smallk reading a channel slice of a wider buffer (the x_buf flow)."""
+    B, H, W = 1, 8, 12
+    buf = torch.randn(B, H, W, 16, device=dev).to(torch.bfloat16)
+    w = torch.randn(32, 2, 7, 7, device=dev) * 0.1
+    bias = torch.zeros(32, device=dev)
+    out = _hip().fconv_smallk(buf.contiguous(), _pack(w), bias, 7, 7, 1,
+                              10, 2)
+    ref = F.relu(F.conv2d(buf[..., 10:12].float().permute(0, 3, 1, 2),
+                          w.to(torch.bfloat16).float(), bias, padding=3))
+    assert (out.float() - ref.permute(0, 2, 3, 1)).abs().max().item() < 0.02
+
+
+def test_fconv_dflow_coords(dev):
+    """delta-flow head with fused coords update == conv + add."""
+    B, H, W, Cin = 1, 7, 11, 64
+    x = torch.randn(B, H, W, Cin, device=dev).to(torch.bfloat16)
+    w = torch.randn(2, Cin, 3, 3, device=dev) * 0.1
+    bias = torch.randn(2, device=dev)
+    coords = torch.randn(B, H, W, 2, device=dev)
+    out = _hip().fconv_dflow_coords(x.contiguous(), _pack(w), bias, coords,
+                                    3, 3)
+    dflow = F.conv2d(x.float().permute(0, 3, 1, 2),
+                     w.to(torch.bfloat16).float(), bias, padding=1)
+    ref = coords + dflow.permute(0, 2, 3, 1)
+    assert (out - ref).abs().max().item() < 0.02
