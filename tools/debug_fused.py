@@ -56,23 +56,29 @@ def main():
         p("convc1", cor_f.permute(0, 3, 1, 2), cor_e)
         cor_f2 = fu.c2(hip, cor_f)
         p("convc2", cor_f2.permute(0, 3, 1, 2), cor_e2)
-        flo_f = fu.f1(hip, flowp)
+        fpad = torch.zeros(B, H8, W8, 8, device=dev, dtype=torch.bfloat16)
+        fpad[..., 0:2] = flowp
+        flo_f = hip.fconv_smallk(fpad, fu.f1.wp, fu.f1.bias, fu.f1.kh,
+                                 fu.f1.kw, fused.ACT_RELU, 0, 2)
         p("convf1", flo_f.permute(0, 3, 1, 2), flo_e)
         flo_f2 = fu.f2(hip, flo_f)
         p("convf2", flo_f2.permute(0, 3, 1, 2), flo_e2)
 
         x_buf = torch.empty(B, H8, W8, 256, device=dev, dtype=torch.bfloat16)
         x_buf[..., :128] = inpp
+        x_buf[..., 254:256] = flowp
         fu.cv(hip, cor_f2, flo_f2, fused.ACT_RELU, out=x_buf, n_off=128)
         p("motion.conv", x_buf[..., 128:254].permute(0, 3, 1, 2), mo_e)
-        x_buf[..., 254:256] = flowp
 
         x2 = torch.empty_like(x_buf)
         x2[..., :128] = inpp
-        newnet, mask, df = fu(hip, netp, x2, corrp, flowp)
+        x2[..., 254:256] = flowp    # the loop's lookup writes this slice
+        coords = torch.randn(B, H8, W8, 2, device=dev)
+        newnet, mask, cnew = fu(hip, netp, x2, corrp, coords)
         p("update.net", newnet.permute(0, 3, 1, 2), ref_net)
         p("update.mask", mask.permute(0, 3, 1, 2), ref_mask)
-        p("update.dflow", df.permute(0, 3, 1, 2), ref_df)
+        p("update.dflow", (cnew - coords).permute(0, 3, 1, 2),
+          ref_df.float())
 
     # full model comparison fused vs eager-bf16 (no fuse)
     x1 = torch.rand(1, 3, 64, 96, device=dev, dtype=torch.bfloat16)
