@@ -64,10 +64,10 @@ class _FC:
         self.wp, self.bias, self.kh, self.kw = pack_conv(conv, pad_cin, scale)
 
     def __call__(self, hip, in1, in2=None, act=ACT_RELU, out=None, n_off=0,
-                 in1_off=0, in1_len=0):
+                 in1_off=0, in1_len=0, stride=1, res=None):
         return hip.fconv_plain(in1, in2, self.wp, self.bias, self.kh,
                                self.kw, act, out, n_off, in1_off, in1_len,
-                               -1, -1)
+                               -1, -1, stride, res)
 
 
 class _GruDir:
@@ -120,7 +120,7 @@ class FusedBasicUpdate:
         # heads: one merged 3x3 conv; mask from a strided slice; the
         # delta-flow final conv also applies coords1 += dflow in-kernel
         hbuf = hip.fconv_plain(net, None, self.heads_w, self.heads_b, 3, 3,
-                               ACT_RELU, None, 0, 0, 0, -1, -1)
+                               ACT_RELU, None, 0, 0, 0, -1, -1, 1, None)
         coords_new = hip.fconv_dflow_coords(hbuf, self.fh2.wp,
                                             self.fh2.bias, coords1, 3, 3)
         mask = self.m2(hip, hbuf, act=ACT_NONE, in1_off=256, in1_len=256)
@@ -157,6 +157,128 @@ class FusedSmallUpdate:
         return net, None, coords_new
 
 
+def pack_raw(w: torch.Tensor, bias, pad_cin: Optional[int] = None):
+    """Pack raw [N, Cin, kh, kw] fp32 weights -> ([taps, N, Cin_pad] bf16,
+    bias fp32)."""
+    N, Cin, kh, kw = w.shape
+    cp = pad_cin if pad_cin is not None else Cin
+    wp = w.new_zeros(N, cp, kh, kw)
+    wp[:, :Cin] = w
+    wp = wp.permute(2, 3, 0, 1).reshape(kh * kw, N, cp)
+    wp = wp.contiguous().to(torch.bfloat16)
+    if bias is None:
+        bias = torch.zeros(N, device=w.device)
+    return wp, bias.contiguous().float(), kh, kw
+
+
+def fold_norm(conv: torch.nn.Conv2d, norm) -> tuple:
+    """Fold an eval-mode BatchNorm (or Identity) into conv weights/bias —
+    cnet norms use running stats at inference (model_utils.py:11)."""
+    w = conv.weight.detach().float()
+    b = conv.bias.detach().float() if conv.bias is not None \
+        else torch.zeros(w.shape[0], device=w.device)
+    if isinstance(norm, torch.nn.BatchNorm2d):
+        scale = norm.weight.detach().float() / torch.sqrt(
+            norm.running_var.detach().float() + norm.eps)
+        w = w * scale.view(-1, 1, 1, 1)
+        b = (b - norm.running_mean.detach().float()) * scale + \
+            norm.bias.detach().float()
+    return w, b
+
+
+class _PC:
+    """Packed conv (optionally norm-folded) with stride/residual support."""
+
+    def __init__(self, conv, norm=None, pad_cin=None, stride=1):
+        w, b = fold_norm(conv, norm) if norm is not None else (
+            conv.weight.detach().float(),
+            conv.bias.detach().float() if conv.bias is not None
+            else torch.zeros(conv.weight.shape[0],
+                             device=conv.weight.device))
+        self.wp, self.bias, self.kh, self.kw = pack_raw(w, b, pad_cin)
+        self.stride = stride
+
+    def __call__(self, hip, x, act=ACT_NONE, res=None):
+        return hip.fconv_plain(x, None, self.wp, self.bias, self.kh,
+                               self.kw, act, None, 0, 0, 0, -1, -1,
+                               self.stride, res)
+
+
+class _FusedResBlock:
+    """ResidualBlock (model_utils.py:19-35) on fconv/inorm kernels.
+    instance norms need the stats kernels; batch/none norms are folded."""
+
+    def __init__(self, blk, norm_fn: str, bottleneck: bool):
+        self.instance = norm_fn == "instance"
+        self.bottleneck = bottleneck
+        s = blk.stride if hasattr(blk, "stride") else \
+            blk.conv2.stride[0] if bottleneck else blk.conv1.stride[0]
+        nf = (lambda m: None) if self.instance else (lambda m: m)
+        self.c1 = _PC(blk.conv1, nf(blk.norm1),
+                      stride=1 if bottleneck else blk.conv1.stride[0])
+        self.c2 = _PC(blk.conv2, nf(blk.norm2), stride=blk.conv2.stride[0])
+        if bottleneck:
+            self.c3 = _PC(blk.conv3, nf(blk.norm3))
+        self.down = None
+        if blk.downsample is not None:
+            self.down = _PC(blk.downsample[0], nf(blk.downsample[1]),
+                            stride=blk.downsample[0].stride[0])
+            self.down_instance = self.instance
+
+    def _norm_relu(self, hip, x, mode=1, res=None):
+        m, r = hip.inorm_stats(x)
+        return hip.inorm_apply(x, m, r, res, mode)
+
+    def __call__(self, hip, x):
+        if self.down is not None:
+            d = self.down(hip, x, ACT_NONE)
+            res = self._norm_relu(hip, d, 0) if self.instance else d
+        else:
+            res = x
+        if self.instance:
+            t = self._norm_relu(hip, self.c1(hip, x))
+            if self.bottleneck:
+                t = self._norm_relu(hip, self.c2(hip, t))
+                y = self.c3(hip, t)
+            else:
+                y = self.c2(hip, t)
+            # relu(res + relu(norm(y))): apply mode 2
+            return self._norm_relu(hip, y, 2, res)
+        # batch (folded) / none
+        t = self.c1(hip, x, ACT_RELU)
+        if self.bottleneck:
+            t = self.c2(hip, t, ACT_RELU)
+            return self.c3(hip, t, ACT_NONE, res=res)   # relu(res+relu(v))
+        return self.c2(hip, t, ACT_NONE, res=res)
+
+
+class FusedEncoder:
+    """Basic/SmallEncoder (model_utils.py:61-105) on the NHWC kernel set.
+    Input: physical-NHWC bf16 images padded to 8 channels."""
+
+    def __init__(self, enc, norm_fn: str):
+        from raft_amd.models.encoders import SmallEncoder
+        bottleneck = isinstance(enc, SmallEncoder)
+        self.instance = norm_fn == "instance"
+        self.conv1 = _PC(enc.conv1,
+                         None if self.instance else enc.norm1,
+                         pad_cin=8, stride=2)
+        self.blocks = []
+        for layer in (enc.layer1, enc.layer2, enc.layer3):
+            for blk in layer:
+                self.blocks.append(_FusedResBlock(blk, norm_fn, bottleneck))
+        self.proj = _PC(enc.conv2)
+
+    def __call__(self, hip, x8):
+        h = self.conv1(hip, x8, ACT_NONE if self.instance else ACT_RELU)
+        if self.instance:
+            m, r = hip.inorm_stats(h)
+            h = hip.inorm_apply(h, m, r, None, 1)
+        for blk in self.blocks:
+            h = blk(hip, h)
+        return self.proj(hip, h, ACT_NONE)
+
+
 class FusedRaft:
     """Caches packed weights for a RAFT model; call run() for inference."""
 
@@ -176,11 +298,18 @@ class FusedRaft:
             self.update = FusedBasicUpdate(model.update_block, self.corr_cpad,
                                            cfg.context_dim)
             self.x_dim = cfg.context_dim + 128
+        import os
+        self.fuse_enc = os.environ.get("RAFT_AMD_EAGER_ENC", "0") != "1"
+        if self.fuse_enc:
+            fnorm = "instance"
+            cnorm = "none" if cfg.small else "batch"
+            self.fnet_f = FusedEncoder(model.fnet, fnorm)
+            self.cnet_f = FusedEncoder(model.cnet, cnorm)
         self._version = self._weights_version()
         self._graphs = {}
 
     def _weights_version(self) -> int:
-        return sum(p._version for p in self.model.update_block.parameters())
+        return sum(p._version for p in self.model.parameters())
 
     def stale(self) -> bool:
         return self._version != self._weights_version()
@@ -224,22 +353,35 @@ class FusedRaft:
         hip = self.hip
         img1 = model.preprocess(image1)
         img2 = model.preprocess(image2)
-        fmaps = model.fnet(torch.cat([img1, img2], dim=0))
-        fmap1, fmap2 = torch.chunk(fmaps, 2, dim=0)
-        # physical NHWC views ([B,C,H,W] channels-last -> permute is free)
-        f1p = fmap1.permute(0, 2, 3, 1).contiguous()
-        f2p = fmap2.permute(0, 2, 3, 1).contiguous()
-        B, H8, W8, C = f1p.shape
 
-        # overlap: the context encoder (MIOpen) runs on a side stream
-        # concurrently with the correlation volume + pyramid build
         side = getattr(self, "_side_stream", None)
         if side is None:
             side = torch.cuda.Stream()
             self._side_stream = side
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            cnet = model.cnet(img1)
+
+        if self.fuse_enc:
+            # in-repo encoders: images -> physical NHWC padded to 8 ch
+            # (Cin=3 would hit the scalar staging seam in every load)
+            B0, _, Hi, Wi = img1.shape
+            x8 = torch.zeros(2 * B0, Hi, Wi, 8, device=img1.device,
+                             dtype=torch.bfloat16)
+            x8[:B0, ..., :3] = img1.permute(0, 2, 3, 1)
+            x8[B0:, ..., :3] = img2.permute(0, 2, 3, 1)
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                cnet_p = self.cnet_f(hip, x8[:B0].contiguous())
+            fmaps_p = self.fnet_f(hip, x8)
+            f1p = fmaps_p[:B0].contiguous()
+            f2p = fmaps_p[B0:].contiguous()
+        else:
+            fmaps = model.fnet(torch.cat([img1, img2], dim=0))
+            fmap1, fmap2 = torch.chunk(fmaps, 2, dim=0)
+            f1p = fmap1.permute(0, 2, 3, 1).contiguous()
+            f2p = fmap2.permute(0, 2, 3, 1).contiguous()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                cnet = model.cnet(img1)
+        B, H8, W8, C = f1p.shape
 
         vol = hip.corr_volume_nhwc(f1p, f2p, True)       # bf16 volume
         levels = [vol]
@@ -251,10 +393,16 @@ class FusedRaft:
                 levels.append(hip.corr_pool2x_bf16(last))
 
         torch.cuda.current_stream().wait_stream(side)
-        cnet.record_stream(torch.cuda.current_stream())
-        net, inp = torch.split(cnet, [cfg.hidden_dim, cfg.context_dim], dim=1)
-        net = torch.tanh(net).permute(0, 2, 3, 1).contiguous()
-        inp = torch.relu(inp).permute(0, 2, 3, 1).contiguous()
+        if self.fuse_enc:
+            cnet_p.record_stream(torch.cuda.current_stream())
+            net = torch.tanh(cnet_p[..., :cfg.hidden_dim]).contiguous()
+            inp = torch.relu(cnet_p[..., cfg.hidden_dim:]).contiguous()
+        else:
+            cnet.record_stream(torch.cuda.current_stream())
+            net, inp = torch.split(cnet, [cfg.hidden_dim, cfg.context_dim],
+                                   dim=1)
+            net = torch.tanh(net).permute(0, 2, 3, 1).contiguous()
+            inp = torch.relu(inp).permute(0, 2, 3, 1).contiguous()
 
         ys, xs = torch.meshgrid(
             torch.arange(H8, device=net.device, dtype=torch.float32),

@@ -46,8 +46,13 @@ void launch_corr_pool2x_bf16(const void*, void*, int, int, int, int,
 void launch_fconv_nhwc_bf16(const void*, int, int, int, const void*, int,
                             const void*, const float*, void*, int, int, int,
                             int, int, int, int, int, int, int, const void*,
-                            const void*, void*, void*, int, int,
+                            const void*, void*, void*, int, int, int,
                             hipStream_t);
+void launch_inorm_stats(const void*, float*, float*, int, int, int, float,
+                        hipStream_t);
+void launch_inorm_apply(const void*, const float*, const float*,
+                        const void*, void*, int, int, int, int,
+                        hipStream_t);
 void launch_fconv_smallk_nhwc_bf16(const void*, int, int, const void*,
                                    const float*, void*, int, int, int, int,
                                    int, int, int, int, hipStream_t);
@@ -294,7 +299,8 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
                        int64_t kh, int64_t kw, int64_t act,
                        c10::optional<at::Tensor> out_buf, int64_t n_off,
                        int64_t in1_off, int64_t in1_len,
-                       int64_t alltaps, int64_t mtiles) {
+                       int64_t alltaps, int64_t mtiles, int64_t stride,
+                       c10::optional<at::Tensor> res) {
     CHECK_DEV(in1); CHECK_CONT(in1); CHECK_DEV(wp); CHECK_CONT(wp);
     TORCH_CHECK(in1.scalar_type() == at::kBFloat16, "fconv needs bf16");
     const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
@@ -316,6 +322,28 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
         TORCH_CHECK(bias->scalar_type() == at::kFloat);
         bptr = bias->data_ptr<float>();
     }
+    if (stride == 1) {
+        TORCH_CHECK((kh == 1 && kw == 1) || (kh == 3 && kw == 3) ||
+                    (kh == 1 && kw == 5) || (kh == 5 && kw == 1),
+                    "fconv: unsupported kernel shape ", kh, "x", kw,
+                    " (use fconv_smallk for tiny-Cin large kernels)");
+    } else {
+        TORCH_CHECK(stride == 2 && kh == kw &&
+                    (kh == 1 || kh == 3 || kh == 7),
+                    "fconv stride-2 supports 1x1/3x3/7x7");
+        TORCH_CHECK(H % 2 == 0 && W % 2 == 0,
+                    "stride-2 fconv needs even input dims");
+    }
+    const int Ho = (stride == 2) ? H / 2 : H;
+    const int Wo = (stride == 2) ? W / 2 : W;
+    const void* res_ptr = nullptr;
+    int mode = 0;
+    if (res.has_value()) {
+        CHECK_CONT(res.value());
+        TORCH_CHECK(res->size(3) == N, "residual channel mismatch");
+        res_ptr = res->data_ptr();
+        mode = 3;   // EP_RES_RELU: out = relu(res + relu(v))
+    }
     at::Tensor out;
     int cstride;
     if (out_buf.has_value()) {
@@ -323,19 +351,16 @@ at::Tensor fconv_plain(at::Tensor in1, c10::optional<at::Tensor> in2,
         CHECK_CONT(out);
         cstride = out.size(3);
     } else {
-        out = at::empty({B, H, W, N}, in1.options());
+        out = at::empty({B, Ho, Wo, N}, in1.options());
         cstride = N;
         n_off = 0;
     }
-    TORCH_CHECK((kh == 1 && kw == 1) || (kh == 3 && kw == 3) ||
-                (kh == 1 && kw == 5) || (kh == 5 && kw == 1),
-                "fconv: unsupported kernel shape ", kh, "x", kw,
-                " (use fconv_smallk for tiny-Cin large kernels)");
     launch_fconv_nhwc_bf16(in1.data_ptr(), C1, in1_stride, (int)in1_off,
                            p2, C2, wp.data_ptr(), bptr, out.data_ptr(), B,
-                           H, W, N, (int)n_off, cstride, (int)kh, (int)kw,
-                           (int)act, 0, nullptr, nullptr, nullptr, nullptr,
-                           (int)alltaps, (int)mtiles, current_stream());
+                           Ho, Wo, N, (int)n_off, cstride, (int)kh, (int)kw,
+                           (int)act, mode, res_ptr, nullptr, nullptr,
+                           nullptr, (int)alltaps, (int)mtiles, (int)stride,
+                           current_stream());
     return out;
 }
 
@@ -352,7 +377,7 @@ std::vector<at::Tensor> fconv_gru_zr(at::Tensor h, at::Tensor x,
                            x.size(3), wp.data_ptr(), bias.data_ptr<float>(),
                            nullptr, B, H, W, N, 0, 0, (int)kh, (int)kw, 0,
                            1, h.data_ptr(), nullptr, z.data_ptr(),
-                           rh.data_ptr(), -1, -1, current_stream());
+                           rh.data_ptr(), -1, -1, 1, current_stream());
     return {z, rh};
 }
 
@@ -369,7 +394,7 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
                            x.size(3), wp.data_ptr(),
                            bias.data_ptr<float>(), out.data_ptr(), B, H, W,
                            hd, 0, hd, (int)kh, (int)kw, 0, 2, h.data_ptr(),
-                           z.data_ptr(), nullptr, nullptr, -1, -1,
+                           z.data_ptr(), nullptr, nullptr, -1, -1, 1,
                            current_stream());
     return out;
 }
@@ -409,6 +434,35 @@ at::Tensor fconv_dflow_coords(at::Tensor in1, at::Tensor wp, at::Tensor bias,
     return out;
 }
 
+std::vector<at::Tensor> inorm_stats(at::Tensor in) {
+    CHECK_DEV(in); CHECK_CONT(in);
+    const int B = in.size(0), H = in.size(1), W = in.size(2);
+    const int C = in.size(3);
+    auto mean = at::empty({B, C}, in.options().dtype(at::kFloat));
+    auto rstd = at::empty({B, C}, in.options().dtype(at::kFloat));
+    launch_inorm_stats(in.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), B, H * W, C, 1e-5f,
+                       current_stream());
+    return {mean, rstd};
+}
+
+at::Tensor inorm_apply(at::Tensor in, at::Tensor mean, at::Tensor rstd,
+                       c10::optional<at::Tensor> res, int64_t mode) {
+    CHECK_DEV(in); CHECK_CONT(in);
+    const int B = in.size(0), H = in.size(1), W = in.size(2);
+    const int C = in.size(3);
+    const void* rptr = nullptr;
+    if (res.has_value()) {
+        CHECK_CONT(res.value());
+        rptr = res->data_ptr();
+    }
+    auto out = at::empty_like(in);
+    launch_inorm_apply(in.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), rptr, out.data_ptr(), B,
+                       H * W, C, (int)mode, current_stream());
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -430,6 +484,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused NHWC bf16 conv (+bias +activation, slice output)");
     m.def("fconv_gru_zr", &fconv_gru_zr, "GRU z/r gate conv pair");
     m.def("fconv_smallk", &fconv_smallk, "direct NHWC conv for tiny Cin");
+    m.def("inorm_stats", &inorm_stats, "per-(b,c) instance-norm stats");
+    m.def("inorm_apply", &inorm_apply,
+          "instance-norm apply (+relu / +residual-relu)");
     m.def("fconv_dflow_coords", &fconv_dflow_coords,
           "flow-head final conv fused with the coords update");
     m.def("fconv_gru_q", &fconv_gru_q, "GRU candidate conv + state update");

@@ -60,6 +60,8 @@ RAFT_DEV float factivate(float v, int act) {
 #define EP_PLAIN 0
 #define EP_GRU_ZR 1   // N = 2*hd: [z | r] -> z_buf = sig(z), rh = sig(r)*h
 #define EP_GRU_Q 2    // N = hd: h' = (1-z)*h + z*tanh(q)
+#define EP_RES_RELU 3 // out = relu(res + relu(v)) — residual blocks
+                      // (res tensor passed via the h_state slot)
 
 // MI/NJ: per-wave 16x16 fragment repeats; tile = (32*MI) x (32*NJ*2)
 // with the fixed 2x2 wave layout. (2,4) = 64x128 (compute-efficient);
@@ -72,7 +74,12 @@ RAFT_DEV float factivate(float v, int act) {
 // position tiles, re-using every staged weight tile MT times (the weight
 // slice is re-staged once per BLOCK per K-step; at batch-1 grids the
 // m-tile count is what multiplies that traffic).
-template <int KH, int KW, int MI, int NJ, bool AT, int MT>
+// S: convolution stride (1 or 2). Stride-2 stages each kernel-row slab as
+// TWO parity sub-slabs (even/odd input columns), so tap reads stay
+// consecutive-row (conflict-free swizzle) — input cols 2m+tx map to
+// parity (tx-PB)&1, row offset (tx-PB)>>1. Requires even input dims
+// (TF-SAME pad (K-2)/2 begin — Conv2dTF semantics for even inputs).
+template <int KH, int KW, int MI, int NJ, bool AT, int MT, int S>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
     int in1_stride, int in1_off,                      // strided slice of in1
@@ -88,9 +95,14 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int TAPS = KH * KW;
     constexpr int BM = 32 * MI;          // block output positions
     constexpr int BN = 32 * NJ;          // block output channels
-    constexpr int AW = BM + KW - 1;      // A slab width (positions + halo)
+    constexpr int PAR = (S == 2) ? 2 : 1;          // parity slabs
+    constexpr int PBW = (S == 2) ? (KW - 2) / 2 : KW / 2;  // left pad
+    constexpr int PBH = (S == 2) ? (KH - 2) / 2 : KH / 2;
+    constexpr int RLO = (S == 2) ? (PBW + 1) / 2 : 0;
+    constexpr int AW = (S == 2) ? (BM + RLO + (KW + 1) / 2 + 1)
+                                : (BM + KW - 1);   // slab rows (+halo)
     constexpr int APAD = ((AW + 15) / 16) * 16;  // swizzle window rounding
-    constexpr int NSLAB = (AT ? KH : 1) * MT;
+    constexpr int NSLAB = (AT ? KH : 1) * MT * PAR;
     constexpr int NBT = AT ? TAPS : KW;
     constexpr int ABYTES = APAD * FC_ROWB;
     constexpr int BBYTES = BN * FC_ROWB;
@@ -131,22 +143,28 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     auto stage = [&](int s) {
         const int ty0 = AT ? 0 : s / ksteps;
         const int k0 = (AT ? s : (s - ty0 * ksteps)) * FC_BK;
-        // ---- A slabs: NSLAB (= rows x m-tiles) x AW x FC_BK ch
+        // ---- A slabs: NSLAB (= rows x m-tiles x parity) x AW x FC_BK ch
+        const int Hi = (S == 2) ? 2 * H : H;     // input dims (even for S2)
+        const int Wi = (S == 2) ? 2 * W : W;
         for (int e = tid; e < NSLAB * AW * (FC_BK / 8); e += 256) {
             const int sl = e / (AW * (FC_BK / 8));
             const int rem0 = e % (AW * (FC_BK / 8));
             const int ar = rem0 / (FC_BK / 8);
             const int c8 = (rem0 % (FC_BK / 8)) * 8;
             char* sA = sAbase + sl * ABYTES;
-            const int mt = sl % MT;              // m-tile index
-            const int rsl = sl / MT;             // kernel-row slab index
-            const int row = y + (ty0 + rsl) - KH / 2;
-            const bool row_ok = (row >= 0 && row < H);
-            const int x = x0 + mt * BM + ar - KW / 2;
+            const int par = sl % PAR;
+            const int mt = (sl / PAR) % MT;      // m-tile index
+            const int rsl = sl / (PAR * MT);     // kernel-row slab index
+            const int row = (S == 2) ? (2 * y + (ty0 + rsl) - PBH)
+                                     : (y + (ty0 + rsl) - PBH);
+            const bool row_ok = (row >= 0 && row < Hi);
+            const int x = (S == 2)
+                ? (2 * (x0 + mt * BM + ar - RLO) + par)
+                : (x0 + mt * BM + ar - PBW);
             uint4v v = {0, 0, 0, 0};
-            if (row_ok && x >= 0 && x < W) {
+            if (row_ok && x >= 0 && x < Wi) {
                 const int k = k0 + c8;
-                const long long p = ((long long)b * H + row) * W + x;
+                const long long p = ((long long)b * Hi + row) * Wi + x;
                 if (k < C1) {
                     if (k + 8 <= C1)
                         v = *(const uint4v*)(in1 + p * in1_stride + in1_off
@@ -211,14 +229,18 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
         for (int rsl = 0; rsl < (AT ? KH : 1); ++rsl) {
 #pragma unroll
             for (int mt = 0; mt < MT; ++mt) {
-                const char* sA = sAbase + (rsl * MT + mt) * ABYTES;
 #pragma unroll
                 for (int tx = 0; tx < KW; ++tx) {
+                    const int d = tx - PBW;
+                    const int par = (S == 2) ? (d & 1) : 0;
+                    const int roff = (S == 2) ? ((d >> 1) + RLO) : (d + PBW);
+                    const char* sA = sAbase +
+                        ((rsl * MT + mt) * PAR + par) * ABYTES;
                     short8 af[MI], bf[NJ];
 #pragma unroll
                     for (int i = 0; i < MI; ++i)
                         af[i] = *(const short8*)(
-                            sA + fswz(wm + i * 16 + (lane & 15) + tx, cb));
+                            sA + fswz(wm + i * 16 + (lane & 15) + roff, cb));
 #pragma unroll
                     for (int j = 0; j < NJ; ++j)
                         bf[j] = *(const short8*)(
@@ -257,6 +279,10 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                 if (mode == EP_PLAIN) {
                     out[p * out_cstride + n_off + n] =
                         (__hip_bfloat16)factivate(v, act);
+                } else if (mode == EP_RES_RELU) {
+                    const float res = (float)h_state[p * hd + n];
+                    out[p * out_cstride + n_off + n] = (__hip_bfloat16)
+                        fmaxf(res + fmaxf(v, 0.0f), 0.0f);
                 } else if (mode == EP_GRU_ZR) {
                     const float s = 1.0f / (1.0f + __expf(-v));
                     if (n < hd) {
@@ -349,7 +375,7 @@ extern "C" void launch_fconv_nhwc_bf16(
     int C2, const void* wp, const float* bias, void* out, int B, int H,
     int W, int N, int n_off, int out_cstride, int kh, int kw, int act,
     int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
-    void* rh_out, int alltaps, int mtiles, hipStream_t s) {
+    void* rh_out, int alltaps, int mtiles, int stride, hipStream_t s) {
     dim3 blk(256);
     const long long big_blocks0 =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
@@ -384,9 +410,35 @@ extern "C" void launch_fconv_nhwc_bf16(
 #define FC_LAUNCH(KH, KW, MI, NJ, AT, MTv, BMv, BNv)                         \
     {                                                                        \
         dim3 grid(cdiv(N, BNv), H * cdiv(W, (BMv) * (MTv)), B);              \
-        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, MI, NJ, AT, MTv>),     \
+        hipLaunchKernelGGL((fconv_nhwc_bf16_k<KH, KW, MI, NJ, AT, MTv, 1>),  \
                            grid, blk, 0, s, FCONV_ARGS);                     \
         return;                                                              \
+    }
+#define FC_LAUNCH2(KH, KW, MI, NJ, BMv, BNv)                                 \
+    {                                                                        \
+        dim3 grid(cdiv(N, BNv), H * cdiv(W, BMv), B);                        \
+        hipLaunchKernelGGL(                                                  \
+            (fconv_nhwc_bf16_k<KH, KW, MI, NJ, false, 1, 2>), grid, blk, 0, \
+            s, FCONV_ARGS);                                                  \
+        return;                                                              \
+    }
+    // stride-2 (encoder) shapes: H/W here are OUTPUT dims; input = 2H x 2W
+    if (stride == 2) {
+        const long long big2 = (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
+        const bool bigt = big2 >= 512;
+        if (kh == 7 && kw == 7) {
+            if (bigt) FC_LAUNCH2(7, 7, 2, 4, 64, 128)
+            FC_LAUNCH2(7, 7, 1, 2, 32, 64)
+        }
+        if (kh == 3 && kw == 3) {
+            if (bigt) FC_LAUNCH2(3, 3, 2, 4, 64, 128)
+            FC_LAUNCH2(3, 3, 1, 2, 32, 64)
+        }
+        if (kh == 1 && kw == 1) {
+            if (bigt) FC_LAUNCH2(1, 1, 2, 4, 64, 128)
+            FC_LAUNCH2(1, 1, 1, 2, 32, 64)
+        }
+        return;  // unsupported stride-2 shape: no-op (binding checks)
     }
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \

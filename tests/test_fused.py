@@ -40,7 +40,7 @@ def test_fconv_matches_conv2d(dev, kh, kw):
     w = torch.randn(N, Cin, kh, kw, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, kh, kw,
-                             1, None, 0, 0, 0, -1, -1)
+                             1, None, 0, 0, 0, -1, -1, 1, None)
     # reference: fp32 conv on the bf16-rounded inputs/weights
     xr = x.float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias,
@@ -59,7 +59,7 @@ def test_fconv_two_inputs_and_slice_output(dev):
     bias = torch.zeros(N, device=dev)
     buf = torch.zeros(B, H, W, 80, device=dev, dtype=torch.bfloat16)
     out = _hip().fconv_plain(a.contiguous(), b.contiguous(), _pack(w), bias,
-                             3, 3, 0, buf, 16, 0, 0, -1, -1)
+                             3, 3, 0, buf, 16, 0, 0, -1, -1, 1, None)
     assert out.data_ptr() == buf.data_ptr()
     xr = torch.cat([a, b], dim=-1).float().permute(0, 3, 1, 2)
     ref = F.conv2d(xr, w.to(torch.bfloat16).float(), bias, padding=1)
@@ -77,7 +77,7 @@ def test_fconv_small_cin_seam(dev):
     w = torch.randn(N, 12, 3, 3, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
     out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, 3, 3,
-                             1, None, 0, 0, 0, -1, -1)
+                             1, None, 0, 0, 0, -1, -1, 1, None)
     ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=1))
     ref = ref.permute(0, 2, 3, 1)
@@ -213,7 +213,7 @@ def test_fconv_strided_input_slice(dev):
     w = torch.randn(24, 32, 3, 3, device=dev) * 0.1
     bias = torch.zeros(24, device=dev)
     out = _hip().fconv_plain(buf.contiguous(), None, _pack(w), bias, 3, 3,
-                             0, None, 0, 40, 32, -1, -1)
+                             0, None, 0, 40, 32, -1, -1, 1, None)
     ref = F.conv2d(buf[..., 40:72].float().permute(0, 3, 1, 2),
                    w.to(torch.bfloat16).float(), bias, padding=1)
     ref = ref.permute(0, 2, 3, 1)
@@ -263,3 +263,85 @@ def test_fconv_dflow_coords(dev):
                      w.to(torch.bfloat16).float(), bias, padding=1)
     ref = coords + dflow.permute(0, 2, 3, 1)
     assert (out - ref).abs().max().item() < 0.02
+
+
+def test_fconv_stride2_matches_conv2dtf(dev):
+    """Stride-2 parity-slab staging vs Conv2dTF (TF-SAME) for 7x7/3x3/1x1."""
+    from raft_amd.models.layers import Conv2dTF
+    for k, cin, n in ((7, 8, 64), (3, 64, 96), (1, 64, 96)):
+        torch.manual_seed(k)
+        conv = Conv2dTF(cin, n, k, stride=2).to(dev)
+        x = torch.randn(1, cin, 64, 96, device=dev)
+        ref = conv(x.to(torch.bfloat16).float())
+        xp = x.to(torch.bfloat16).permute(0, 2, 3, 1).contiguous()
+        w = _pack(conv.weight.detach().float())
+        out = _hip().fconv_plain(xp, None, w,
+                                 conv.bias.detach().float().contiguous(),
+                                 k, k, 0, None, 0, 0, 0, -1, -1, 2, None)
+        got = out.float().permute(0, 3, 1, 2)
+        assert got.shape == ref.shape, (k, got.shape, ref.shape)
+        err = (got - ref).abs().max().item()
+        assert err < 0.05 * ref.abs().max().item() + 0.05, (k, err)
+
+
+def test_fconv_residual_epilogue(dev):
+    """EP_RES_RELU: out = relu(res + relu(conv(x)))."""
+    B, H, W, C, N = 1, 8, 12, 32, 32
+    x = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
+    res = torch.randn(B, H, W, N, device=dev).to(torch.bfloat16)
+    w = torch.randn(N, C, 3, 3, device=dev) * 0.1
+    bias = torch.zeros(N, device=dev)
+    out = _hip().fconv_plain(x.contiguous(), None, _pack(w), bias, 3, 3, 0,
+                             None, 0, 0, 0, -1, -1, 1, res.contiguous())
+    y = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
+                        w.to(torch.bfloat16).float(), bias, padding=1))
+    ref = F.relu(res.float().permute(0, 3, 1, 2) + y).permute(0, 2, 3, 1)
+    assert (out.float() - ref).abs().max().item() < 0.05
+
+
+def test_inorm_kernels(dev):
+    B, H, W, C = 2, 9, 13, 96
+    x = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
+    m, r = _hip().inorm_stats(x.contiguous())
+    xf = x.float()
+    ref_m = xf.mean(dim=(1, 2))
+    ref_v = xf.var(dim=(1, 2), unbiased=False)
+    assert torch.allclose(m, ref_m, atol=1e-3)
+    assert torch.allclose(r, 1.0 / torch.sqrt(ref_v + 1e-5), atol=1e-2,
+                          rtol=1e-2)
+    out = _hip().inorm_apply(x.contiguous(), m, r, None, 1)
+    ref = F.relu((xf - ref_m[:, None, None]) /
+                 torch.sqrt(ref_v[:, None, None] + 1e-5))
+    assert (out.float() - ref).abs().max().item() < 0.02
+    # mode 2 with residual
+    res = torch.randn_like(x)
+    out2 = _hip().inorm_apply(x.contiguous(), m, r, res.contiguous(), 2)
+    ref2 = F.relu(res.float() + ref)
+    assert (out2.float() - ref2).abs().max().item() < 0.03
+
+
+def test_fused_encoder_matches_eager(dev):
+    """FusedEncoder vs the eager (bf16, channels-last) encoders."""
+    from raft_amd.models.encoders import BasicEncoder, SmallEncoder
+    from raft_amd.models.fused import FusedEncoder
+    cases = [
+        (BasicEncoder(output_dim=256, norm_fn="instance"), "instance"),
+        (BasicEncoder(output_dim=256, norm_fn="batch"), "batch"),
+        (SmallEncoder(output_dim=128, norm_fn="instance"), "instance"),
+        (SmallEncoder(output_dim=160, norm_fn="none"), "none"),
+    ]
+    x = torch.rand(2, 3, 64, 96, device=dev) * 2 - 1
+    for enc, nf in cases:
+        enc = enc.to(dev).to(torch.bfloat16).eval() \
+            .to(memory_format=torch.channels_last)
+        with torch.no_grad():
+            ref = enc(x.to(torch.bfloat16)
+                      .contiguous(memory_format=torch.channels_last))
+            fe = FusedEncoder(enc, nf)
+            x8 = torch.zeros(2, 64, 96, 8, device=dev, dtype=torch.bfloat16)
+            x8[..., :3] = x.to(torch.bfloat16).permute(0, 2, 3, 1)
+            got = fe(_hip(), x8).permute(0, 3, 1, 2)
+        refp = ref.float()
+        err = (got.float() - refp).abs().max().item()
+        scale = refp.abs().max().item()
+        assert err < 0.05 * scale + 0.05, (type(enc).__name__, nf, err)
