@@ -8,20 +8,24 @@
 #include "common.h"
 #include <hip/hip_bf16.h>
 
-// stats: grid (B, ceil(C/64)); block 256 = 64 channels x 4 row-groups
-extern "C" __global__ __launch_bounds__(256) void inorm_stats_k(
+// stats, pass 1: split reduction — grid (B, ceil(C/64), S position
+// partitions); each block reduces its partition into LDS then one
+// atomicAdd per (c, moment) into the zero-initialized accumulator.
+// (A (B, C/64) grid was 2 blocks on a 256-CU chip: 3.2 ms/call.)
+extern "C" __global__ __launch_bounds__(256) void inorm_stats_part_k(
     const __hip_bfloat16* __restrict__ in,   // [B, H*W, C]
-    float* __restrict__ mean, float* __restrict__ rstd,  // [B, C]
-    int HW, int C, float eps) {
+    float* __restrict__ acc,                 // [B, C, 2] zero-initialized
+    int HW, int C) {
     __shared__ float red[2][4][64];
     const int b = blockIdx.x;
     const int c0 = blockIdx.y * 64;
     const int c = c0 + (threadIdx.x & 63);
     const int g = threadIdx.x >> 6;          // row-group 0..3
+    const int S = gridDim.z;
     float s = 0.f, s2 = 0.f;
     if (c < C) {
         const __hip_bfloat16* base = in + (size_t)b * HW * C + c;
-        for (int p = g; p < HW; p += 4) {
+        for (int p = blockIdx.z * 4 + g; p < HW; p += 4 * S) {
             const float v = (float)base[(size_t)p * C];
             s += v;
             s2 += v * v;
@@ -36,11 +40,21 @@ extern "C" __global__ __launch_bounds__(256) void inorm_stats_k(
             ts += red[0][i][threadIdx.x & 63];
             ts2 += red[1][i][threadIdx.x & 63];
         }
-        const float m = ts / HW;
-        const float var = fmaxf(ts2 / HW - m * m, 0.0f);
-        mean[(size_t)b * C + c] = m;
-        rstd[(size_t)b * C + c] = rsqrtf(var + eps);
+        atomicAdd(&acc[((size_t)b * C + c) * 2], ts);
+        atomicAdd(&acc[((size_t)b * C + c) * 2 + 1], ts2);
     }
+}
+
+// stats, pass 2: finalize mean/rstd from the accumulated moments
+extern "C" __global__ void inorm_stats_fin_k(
+    const float* __restrict__ acc, float* __restrict__ mean,
+    float* __restrict__ rstd, int HW, long long BC, float eps) {
+    const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= BC) return;
+    const float m = acc[i * 2] / HW;
+    const float var = fmaxf(acc[i * 2 + 1] / HW - m * m, 0.0f);
+    mean[i] = m;
+    rstd[i] = rsqrtf(var + eps);
 }
 
 // apply: y = act((x - mean) * rstd) [+ residual, outer relu]
@@ -65,12 +79,17 @@ extern "C" __global__ void inorm_apply_k(
     }
 }
 
-extern "C" void launch_inorm_stats(const void* in, float* mean, float* rstd,
-                                   int B, int HW, int C, float eps,
-                                   hipStream_t s) {
-    dim3 grid(B, cdiv(C, 64));
-    hipLaunchKernelGGL(inorm_stats_k, grid, dim3(256), 0, s,
-                       (const __hip_bfloat16*)in, mean, rstd, HW, C, eps);
+extern "C" void launch_inorm_stats(const void* in, float* acc, float* mean,
+                                   float* rstd, int B, int HW, int C,
+                                   float eps, hipStream_t s) {
+    const int S = (int)min((long long)cdiv(HW, 1024), (long long)128);
+    dim3 grid(B, cdiv(C, 64), S);
+    hipLaunchKernelGGL(inorm_stats_part_k, grid, dim3(256), 0, s,
+                       (const __hip_bfloat16*)in, acc, HW, C);
+    const long long BC = (long long)B * C;
+    hipLaunchKernelGGL(inorm_stats_fin_k,
+                       dim3((unsigned)((BC + 255) / 256)), dim3(256), 0, s,
+                       acc, mean, rstd, HW, BC, eps);
 }
 
 extern "C" void launch_inorm_apply(const void* in, const float* mean,
