@@ -111,7 +111,7 @@ class FusedBasicUpdate:
         cor = self.c2(hip, self.c1(hip, corr_pad))
         flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
                                 self.f1.kh, self.f1.kw, ACT_RELU,
-                                ctx + 126, 2)
+                                ctx + 126, 2, 1)
         flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
         # SepConvGRU (model_utils.py:138-156)
@@ -147,7 +147,7 @@ class FusedSmallUpdate:
         cor = self.c1(hip, corr_pad)
         flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
                                 self.f1.kh, self.f1.kw, ACT_RELU,
-                                ctx + 80, 2)
+                                ctx + 80, 2, 1)
         flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
         net = self.gru(hip, net, x_buf)
@@ -260,9 +260,12 @@ class FusedEncoder:
         from raft_amd.models.encoders import SmallEncoder
         bottleneck = isinstance(enc, SmallEncoder)
         self.instance = norm_fn == "instance"
-        self.conv1 = _PC(enc.conv1,
-                         None if self.instance else enc.norm1,
-                         pad_cin=8, stride=2)
+        # stem 7x7/2 on 3 channels: small-K direct kernel (an MFMA tile
+        # would spend 29/32 of its K on zero padding — measured 409 us)
+        w1, b1 = fold_norm(enc.conv1,
+                           None if self.instance else enc.norm1) \
+            if not self.instance else fold_norm(enc.conv1, None)
+        self.conv1_w, self.conv1_b, self.c1kh, self.c1kw = pack_raw(w1, b1)
         self.blocks = []
         for layer in (enc.layer1, enc.layer2, enc.layer3):
             for blk in layer:
@@ -270,7 +273,10 @@ class FusedEncoder:
         self.proj = _PC(enc.conv2)
 
     def __call__(self, hip, x8):
-        h = self.conv1(hip, x8, ACT_NONE if self.instance else ACT_RELU)
+        h = hip.fconv_smallk(x8, self.conv1_w, self.conv1_b, self.c1kh,
+                             self.c1kw,
+                             ACT_NONE if self.instance else ACT_RELU,
+                             0, 3, 2)
         if self.instance:
             m, r = hip.inorm_stats(h)
             h = hip.inorm_apply(h, m, r, None, 1)

@@ -55,7 +55,7 @@ void launch_inorm_apply(const void*, const float*, const float*,
                         hipStream_t);
 void launch_fconv_smallk_nhwc_bf16(const void*, int, int, const void*,
                                    const float*, void*, int, int, int, int,
-                                   int, int, int, int, hipStream_t);
+                                   int, int, int, int, int, hipStream_t);
 void launch_fconv_dflow_coords(const void*, int, int, int, const void*,
                                const float*, const float*, float*, int, int,
                                int, int, int, hipStream_t);
@@ -402,18 +402,21 @@ at::Tensor fconv_gru_q(at::Tensor rh, at::Tensor x, at::Tensor wp,
 at::Tensor fconv_smallk(at::Tensor in1, at::Tensor wp,
                         c10::optional<at::Tensor> bias, int64_t kh,
                         int64_t kw, int64_t act, int64_t in1_off,
-                        int64_t in1_len) {
+                        int64_t in1_len, int64_t conv_stride) {
     CHECK_DEV(in1); CHECK_CONT(in1); CHECK_CONT(wp);
     const int B = in1.size(0), H = in1.size(1), W = in1.size(2);
     const int stride = in1.size(3);
     const int C = in1_len > 0 ? (int)in1_len : stride;
     const int N = wp.size(1);
     const float* bptr = bias.has_value() ? bias->data_ptr<float>() : nullptr;
-    auto out = at::empty({B, H, W, N}, in1.options());
+    const int Ho = (int)(H / conv_stride), Wo = (int)(W / conv_stride);
+    TORCH_CHECK(conv_stride == 1 ||
+                (conv_stride == 2 && H % 2 == 0 && W % 2 == 0));
+    auto out = at::empty({B, Ho, Wo, N}, in1.options());
     launch_fconv_smallk_nhwc_bf16(in1.data_ptr(), stride, (int)in1_off,
-                                  wp.data_ptr(), bptr, out.data_ptr(), B, H,
-                                  W, C, N, (int)kh, (int)kw, (int)act,
-                                  current_stream());
+                                  wp.data_ptr(), bptr, out.data_ptr(), B,
+                                  Ho, Wo, C, N, (int)kh, (int)kw, (int)act,
+                                  (int)conv_stride, current_stream());
     return out;
 }
 

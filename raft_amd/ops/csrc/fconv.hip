@@ -470,12 +470,13 @@ extern "C" void launch_fconv_nhwc_bf16(
 #define SK_MAXIN (8 * 49 * 4)        // PPB x taps x C cap
 
 extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
-    const __hip_bfloat16* __restrict__ in,    // [B, H, W, *] slice
+    const __hip_bfloat16* __restrict__ in,    // [B, Hi, Wi, *] slice
     int in_stride, int in_off,
     const __hip_bfloat16* __restrict__ wp,    // [kh*kw][N][C]
     const float* __restrict__ bias,
-    __hip_bfloat16* __restrict__ out,         // [B, H, W, N]
-    int H, int W, int C, int N, int kh, int kw, int act, long long ncells) {
+    __hip_bfloat16* __restrict__ out,         // [B, H, W, N] (output dims)
+    int H, int W, int C, int N, int kh, int kw, int act, int cstride,
+    long long ncells) {
     __shared__ __hip_bfloat16 smem[SK_MAXW + SK_MAXIN];
     __hip_bfloat16* sw = smem;
     __hip_bfloat16* sin = smem + SK_MAXW;
@@ -499,10 +500,16 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
                 const int x = (int)(cell % W);
                 const int y = (int)((cell / W) % H);
                 const int b = (int)(cell / ((long long)W * H));
-                const int yy = y + t / kw - kh / 2;
-                const int xx = x + t % kw - kw / 2;
-                if (yy >= 0 && yy < H && xx >= 0 && xx < W)
-                    v = in[(((long long)b * H + yy) * W + xx) * in_stride
+                // cstride == 2: TF-SAME stride-2 (pad begin (K-2)/2,
+                // even input dims = 2H x 2W); else stride-1 SAME
+                const int Hi = cstride * H;
+                const int Wi = cstride * W;
+                const int pb = (cstride == 2) ? (kh - 2) / 2 : kh / 2;
+                const int pbw = (cstride == 2) ? (kw - 2) / 2 : kw / 2;
+                const int yy = cstride * y + t / kw - pb;
+                const int xx = cstride * x + t % kw - pbw;
+                if (yy >= 0 && yy < Hi && xx >= 0 && xx < Wi)
+                    v = in[(((long long)b * Hi + yy) * Wi + xx) * in_stride
                            + in_off + c];
             }
             sin[e] = v;
@@ -562,7 +569,8 @@ extern "C" __global__ void fconv_smallk_nhwc_bf16_k(
 extern "C" void launch_fconv_smallk_nhwc_bf16(
     const void* in, int in_stride, int in_off, const void* wp,
     const float* bias, void* out, int B, int H, int W, int C, int N, int kh,
-    int kw, int act, hipStream_t s) {
+    int kw, int act, int conv_stride, hipStream_t s) {
+    // H, W are OUTPUT dims (input = conv_stride*H x conv_stride*W)
     const long long ncells = (long long)B * H * W;
     const int taps = kh * kw;
     if (N <= 128 && 256 % N == 0 && taps * N * C <= SK_MAXW &&
@@ -578,9 +586,11 @@ extern "C" void launch_fconv_smallk_nhwc_bf16(
                            s, (const __hip_bfloat16*)in, in_stride, in_off,
                            (const __hip_bfloat16*)wp, bias,
                            (__hip_bfloat16*)out, H, W, C, N, kh, kw, act,
-                           ncells);
+                           conv_stride, ncells);
         return;
     }
+    // naive fallback is stride-1 only
+    if (conv_stride != 1) return;
     const long long total = ncells * N;
     int blocks = (int)min((total + 255) / 256, (long long)4096);
     hipLaunchKernelGGL(fconv_smallk_nhwc_bf16_k, dim3(blocks), dim3(256), 0,

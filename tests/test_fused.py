@@ -180,7 +180,7 @@ def test_fconv_smallk_matches_conv2d(dev):
     x = torch.randn(B, H, W, C, device=dev).to(torch.bfloat16)
     w = torch.randn(N, C, 7, 7, device=dev) * 0.1
     bias = torch.randn(N, device=dev)
-    out = _hip().fconv_smallk(x.contiguous(), _pack(w), bias, 7, 7, 1, 0, 0)
+    out = _hip().fconv_smallk(x.contiguous(), _pack(w), bias, 7, 7, 1, 0, 0, 1)
     ref = F.relu(F.conv2d(x.float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=3))
     ref = ref.permute(0, 2, 3, 1)
@@ -244,7 +244,7 @@ def test_fconv_smallk_slice_input(dev):
     w = torch.randn(32, 2, 7, 7, device=dev) * 0.1
     bias = torch.zeros(32, device=dev)
     out = _hip().fconv_smallk(buf.contiguous(), _pack(w), bias, 7, 7, 1,
-                              10, 2)
+                              10, 2, 1)
     ref = F.relu(F.conv2d(buf[..., 10:12].float().permute(0, 3, 1, 2),
                           w.to(torch.bfloat16).float(), bias, padding=3))
     assert (out.float() - ref.permute(0, 2, 3, 1)).abs().max().item() < 0.02
@@ -345,3 +345,20 @@ def test_fused_encoder_matches_eager(dev):
         err = (got.float() - refp).abs().max().item()
         scale = refp.abs().max().item()
         assert err < 0.05 * scale + 0.05, (type(enc).__name__, nf, err)
+
+
+def test_fconv_smallk_stride2_stem(dev):
+    """Stem shape: 7x7 stride-2 on 3 channels vs Conv2dTF."""
+    from raft_amd.models.layers import Conv2dTF
+    conv = Conv2dTF(3, 32, 7, stride=2).to(dev)
+    x = torch.rand(2, 3, 64, 96, device=dev)
+    ref = conv(x.to(torch.bfloat16).float())
+    x8 = torch.zeros(2, 64, 96, 8, device=dev, dtype=torch.bfloat16)
+    x8[..., :3] = x.to(torch.bfloat16).permute(0, 2, 3, 1)
+    out = _hip().fconv_smallk(x8.contiguous(),
+                              _pack(conv.weight.detach().float()),
+                              conv.bias.detach().float().contiguous(),
+                              7, 7, 0, 0, 3, 2)
+    got = out.float().permute(0, 3, 1, 2)
+    assert got.shape == ref.shape
+    assert (got - ref).abs().max().item() < 0.05 * ref.abs().max().item() + 0.05
