@@ -211,8 +211,6 @@ class _FusedResBlock:
     def __init__(self, blk, norm_fn: str, bottleneck: bool):
         self.instance = norm_fn == "instance"
         self.bottleneck = bottleneck
-        s = blk.stride if hasattr(blk, "stride") else \
-            blk.conv2.stride[0] if bottleneck else blk.conv1.stride[0]
         nf = (lambda m: None) if self.instance else (lambda m: m)
         self.c1 = _PC(blk.conv1, nf(blk.norm1),
                       stride=1 if bottleneck else blk.conv1.stride[0])
@@ -477,6 +475,7 @@ class FusedRaft:
                                    st["coords0"], st["coords1"], iters)
         return graph, st
 
+
 def get_fused(model) -> Optional[FusedRaft]:
     """Return (building if needed) the packed fused runner for this model.
     Rebuilds automatically when the underlying weights changed."""
@@ -497,5 +496,7 @@ def can_fuse(model, image1: torch.Tensor) -> bool:
         p = next(model.update_block.parameters())
     except StopIteration:
         return False
+    cfg = model.cfg
     return (image1.is_cuda and not torch.is_grad_enabled()
-            and p.dtype == torch.bfloat16 and O.hip_available())
+            and p.dtype == torch.bfloat16 and O.hip_available()
+            and cfg.corr_levels <= 4 and cfg.fnet_dim % 64 == 0)

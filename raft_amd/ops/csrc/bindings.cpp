@@ -250,6 +250,7 @@ at::Tensor corr_lookup_nhwc(std::vector<at::Tensor> levels,
     // coords: [B, H, W, 2] fp32; returns physical NHWC [B, H, W, c_stride]
     // with channels [L*KK..c_stride) zero-filled (pad for fconv Cin%8).
     CHECK_DEV(coords); CHECK_CONT(coords);
+    TORCH_CHECK(!levels.empty() && levels.size() <= 4, "1..4 pyramid levels");
     const int B = coords.size(0), H = coords.size(1), W = coords.size(2);
     const int L = (int)levels.size();
     const int K = 2 * (int)radius + 1;
