@@ -1,0 +1,51 @@
+"""Driver-contract tests for bench.py: single-rank CPU fallback JSON and a
+torchrun-style world-2 gloo run (the exact invocation shape the round-end
+scaling harness uses)."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _parse_last_json(out: str) -> dict:
+    lines = [l for l in out.strip().splitlines() if l.startswith("{")]
+    assert lines, out
+    return json.loads(lines[-1])
+
+
+def test_bench_single_rank_contract():
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    j = _parse_last_json(r.stdout)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in j, key
+    assert j["n_gpus"] == 1 and j["data"] == "synthetic"
+    assert j["scaling"] == "weak" and j["higher_is_better"] is True
+    assert j["value"] > 0
+
+
+def test_bench_torchrun_world2_gloo():
+    """The driver launches N>1 via torch.distributed.run; verify the
+    distributed path end to end on CPU (gloo): exactly one JSON line, from
+    rank 0, with n_gpus == world size."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29731", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1"],
+        cwd=REPO, capture_output=True, text=True, timeout=900, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    jsons = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(jsons) == 1, jsons
+    j = json.loads(jsons[0])
+    assert j["n_gpus"] == 2
+    assert j["config"]["parallelism"] == "dp2"
