@@ -477,44 +477,57 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
     __hip_bfloat16* __restrict__ out,         // [B, H, W, N] (output dims)
     int H, int W, int C, int N, int kh, int kw, int act, int cstride,
     long long ncells) {
-    // weights staged ONCE per block; after that a barrier-free grid-stride
-    // loop over cells. Input windows are read straight from global: the
-    // 64 lanes of one cell share identical addresses (L1 broadcast), so
-    // LDS window staging (and its per-group barriers, which serialized
-    // this kernel at ~1000+ cycles/group) buys nothing.
-    __shared__ __hip_bfloat16 sw[SK_MAXW];
+    __shared__ __hip_bfloat16 smem[SK_MAXW + SK_MAXIN];
+    __hip_bfloat16* sw = smem;
+    __hip_bfloat16* sin = smem + SK_MAXW;
     const int taps = kh * kw;
+    const int PPB = 256 / N;
     const int tid = threadIdx.x;
     for (int e = tid; e < taps * N * C; e += 256) sw[e] = wp[e];
-    __syncthreads();
 
-    const int PPB = 256 / N;
-    const int p = tid / N;
-    const int n = tid - p * N;
-    const int Hi = cstride * H;
-    const int Wi = cstride * W;
-    const int pb = (cstride == 2) ? (kh - 2) / 2 : kh / 2;
-    const int pbw = (cstride == 2) ? (kw - 2) / 2 : kw / 2;
     const long long groups = (ncells + PPB - 1) / PPB;
     for (long long g = blockIdx.x; g < groups; g += gridDim.x) {
-        const long long cell = g * (long long)PPB + p;
-        if (cell >= ncells) continue;
-        const int x = (int)(cell % W);
-        const int y = (int)((cell / W) % H);
-        const int b = (int)(cell / ((long long)W * H));
-        float acc = bias ? bias[n] : 0.0f;
-        for (int t = 0; t < taps; ++t) {
-            const int yy = cstride * y + t / kw - pb;
-            const int xx = cstride * x + t % kw - pbw;
-            if (yy < 0 || yy >= Hi || xx < 0 || xx >= Wi) continue;
-            const __hip_bfloat16* src =
-                in + (((long long)b * Hi + yy) * Wi + xx) * in_stride
-                + in_off;
-            const __hip_bfloat16* wr = sw + ((size_t)t * N + n) * C;
-            for (int c = 0; c < C; ++c)
-                acc = fmaf((float)src[c], (float)wr[c], acc);
+        const long long cell0 = g * (long long)PPB;
+        // stage PPB windows (zero-padded SAME)
+        for (int e = tid; e < PPB * taps * C; e += 256) {
+            const int p = e / (taps * C);
+            const int rem = e % (taps * C);
+            const int t = rem / C;
+            const int c = rem % C;
+            const long long cell = cell0 + p;
+            __hip_bfloat16 v = (__hip_bfloat16)0.f;
+            if (cell < ncells) {
+                const int x = (int)(cell % W);
+                const int y = (int)((cell / W) % H);
+                const int b = (int)(cell / ((long long)W * H));
+                // cstride == 2: TF-SAME stride-2 (pad begin (K-2)/2,
+                // even input dims = 2H x 2W); else stride-1 SAME
+                const int Hi = cstride * H;
+                const int Wi = cstride * W;
+                const int pb = (cstride == 2) ? (kh - 2) / 2 : kh / 2;
+                const int pbw = (cstride == 2) ? (kw - 2) / 2 : kw / 2;
+                const int yy = cstride * y + t / kw - pb;
+                const int xx = cstride * x + t % kw - pbw;
+                if (yy >= 0 && yy < Hi && xx >= 0 && xx < Wi)
+                    v = in[(((long long)b * Hi + yy) * Wi + xx) * in_stride
+                           + in_off + c];
+            }
+            sin[e] = v;
         }
-        out[cell * N + n] = (__hip_bfloat16)factivate(acc, act);
+        __syncthreads();
+        const int p = tid / N;
+        const int n = tid - p * N;
+        const long long cell = cell0 + p;
+        if (cell < ncells) {
+            float acc = bias ? bias[n] : 0.0f;
+            const __hip_bfloat16* win = sin + p * taps * C;
+            for (int t = 0; t < taps; ++t)
+                for (int c = 0; c < C; ++c)
+                    acc = fmaf((float)win[t * C + c],
+                               (float)sw[(t * N + n) * C + c], acc);
+            out[cell * N + n] = (__hip_bfloat16)factivate(acc, act);
+        }
+        __syncthreads();
     }
 }
 
@@ -564,7 +577,11 @@ extern "C" void launch_fconv_smallk_nhwc_bf16(
         (256 / N) * taps * C <= SK_MAXIN) {
         const int PPB = 256 / N;
         const long long groups = (ncells + PPB - 1) / PPB;
-        int blocks = (int)min(groups, (long long)2048);
+        // cap the grid well below the group count: every block stages the
+        // full weight set once (25 KB for convf1) — 3520 one-group blocks
+        // cost 88 MB of weight reads (measured 43.5 us); ~640 blocks
+        // amortize it ~6x while still filling 256 CUs.
+        int blocks = (int)min(groups, (long long)640);
         hipLaunchKernelGGL(fconv_smallk_lds_k, dim3(blocks), dim3(256), 0,
                            s, (const __hip_bfloat16*)in, in_stride, in_off,
                            (const __hip_bfloat16*)wp, bias,
