@@ -54,6 +54,9 @@ def parse_args(argv=None):
     p.add_argument("--steps", type=int, default=100, help="training steps")
     p.add_argument("--no-graph", action="store_true",
                    help="disable HIP-graph capture in test mode")
+    p.add_argument("--warm", action="store_true",
+                   help="warm-start each sequence pair from the previous "
+                        "flow (official-RAFT 2-view style)")
     return p.parse_args(argv)
 
 
@@ -106,9 +109,18 @@ def mode_test(args, device):
     ds = PairDataflow(pairs, input_size=size, batch=args.batch)
     os.makedirs(args.out, exist_ok=True)
     variant = "raft-small" if args.small else "raft-things"
+    prev_flow = None
     for i, (im1, im2) in enumerate(ds):
+        flow_init = None
+        if args.warm and prev_flow is not None:
+            import torch.nn.functional as Fn
+            h8, w8 = prev_flow.shape[-2] // 8, prev_flow.shape[-1] // 8
+            flow_init = Fn.interpolate(prev_flow, size=(h8, w8),
+                                       mode="bilinear",
+                                       align_corners=False) / 8.0
         t0 = time.perf_counter()
-        flow = engine(im1, im2)
+        flow = engine(im1, im2, flow_init=flow_init)
+        prev_flow = flow.float()
         if device.type == "cuda":
             torch.cuda.synchronize()
         dt = time.perf_counter() - t0

@@ -68,7 +68,10 @@ class InferenceEngine:
 
     @torch.no_grad()
     def __call__(self, image1: torch.Tensor, image2: torch.Tensor,
-                 iters: Optional[int] = None) -> torch.Tensor:
+                 iters: Optional[int] = None,
+                 flow_init: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """flow_init: [B,2,H/8,W/8] warm-start at 1/8 resolution (official
+        RAFT 2-view warm start; the BASELINE EPE anchor is quoted with it)."""
         iters = iters if iters is not None else self.iters
         image1 = image1.to(self.device, self.dtype, non_blocking=True)
         image2 = image2.to(self.device, self.dtype, non_blocking=True)
@@ -83,9 +86,11 @@ class InferenceEngine:
             # measured slower than eager on this stack); opt in with
             # loop_graph=True
             self.model._fused_use_graph = self.loop_graph
-            return unpad(self.model(image1, image2, iters=iters), hw)
-        if not self.use_graph:
-            return unpad(self.model(image1, image2, iters=iters), hw)
+            return unpad(self.model(image1, image2, iters=iters,
+                                    flow_init=flow_init), hw)
+        if not self.use_graph or flow_init is not None:
+            return unpad(self.model(image1, image2, iters=iters,
+                                    flow_init=flow_init), hw)
 
         key = (image1.shape[0], image1.shape[2], image1.shape[3],
                iters if iters is not None else -1)

@@ -127,3 +127,15 @@ def test_cli_test_mode_sequence_dir(tmp_path):
                      "--out", str(out), "--size", "40x56", "--iters", "2"])
     assert (out / "raft_flow_raft-small_0000.png").exists()
     assert (out / "raft_flow_raft-small_0001.flo").exists()
+
+
+def test_engine_warm_start():
+    from raft_amd import RAFT, RaftConfig
+    eng = InferenceEngine(RAFT(RaftConfig(small=True)), iters=2)
+    x1 = torch.rand(1, 3, 64, 96)
+    x2 = torch.rand(1, 3, 64, 96)
+    cold = eng(x1, x2)
+    warm = eng(x1, x2, flow_init=torch.zeros(1, 2, 8, 12))
+    assert torch.allclose(cold, warm, atol=1e-5)   # zero init == cold
+    warm2 = eng(x1, x2, flow_init=torch.ones(1, 2, 8, 12))
+    assert not torch.allclose(cold, warm2)
