@@ -362,3 +362,25 @@ def test_fconv_smallk_stride2_stem(dev):
     got = out.float().permute(0, 3, 1, 2)
     assert got.shape == ref.shape
     assert (got - ref).abs().max().item() < 0.05 * ref.abs().max().item() + 0.05
+
+
+@pytest.mark.parametrize("h,w", [(40, 56), (48, 104), (64, 64), (96, 40)])
+def test_fused_model_shape_sweep(dev, h, w):
+    """Edge shapes (odd tile fractions, W not multiple of 64/32) through
+    the full fused path vs eager bf16."""
+    import os as _os
+
+    from raft_amd import RAFT, RaftConfig
+    torch.manual_seed(5)
+    m = RAFT(RaftConfig(small=False)).to(dev).to(torch.bfloat16).eval()
+    x1 = torch.rand(1, 3, h, w, device=dev, dtype=torch.bfloat16)
+    x2 = torch.rand(1, 3, h, w, device=dev, dtype=torch.bfloat16)
+    with torch.no_grad():
+        _os.environ["RAFT_AMD_NO_FUSE"] = "1"
+        try:
+            ref = m(x1, x2, iters=3)
+        finally:
+            _os.environ.pop("RAFT_AMD_NO_FUSE")
+        out = m(x1, x2, iters=3)
+    err = (out.float() - ref.float()).abs().max().item()
+    assert err < 0.05, (h, w, err)
