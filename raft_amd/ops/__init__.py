@@ -25,14 +25,33 @@ if os.environ.get("RAFT_AMD_FORCE_TORCH", "0") != "1":
         _hip_import_error = e
 
 
+def _try_bind():
+    """(Re)try binding the extension — the .so may have been built after
+    this package was first imported (e.g. __graft_entry__.build() in the
+    same process)."""
+    global _hip_ops, _hip_import_error
+    if _hip_ops is not None:
+        return _hip_ops
+    if os.environ.get("RAFT_AMD_FORCE_TORCH", "0") == "1":
+        return None
+    try:
+        import importlib
+        importlib.invalidate_caches()
+        _hip_ops = importlib.import_module("raft_amd.ops._hip_ops")
+        _hip_import_error = None
+    except ImportError as e:
+        _hip_import_error = e
+    return _hip_ops
+
+
 def hip_available() -> bool:
     """True when the HIP extension is importable AND a GPU is present."""
-    return _hip_ops is not None and torch.cuda.is_available()
+    return _try_bind() is not None and torch.cuda.is_available()
 
 
 def require_hip():
     """Return the HIP extension module, failing loudly if missing on GPU."""
-    if _hip_ops is None:
+    if _try_bind() is None:
         raise RuntimeError(
             "raft_amd HIP extension (_hip_ops) is not built but a GPU run "
             "requested it. Build in-tree with "
