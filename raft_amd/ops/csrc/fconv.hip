@@ -196,28 +196,43 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             }
             *(uint4v*)(sA + fswz(ar, c8 * 2)) = v;
         }
-        // ---- weight tiles [NBT][BN][FC_BK]
-        for (int e = tid; e < NBT * BN * (FC_BK / 8); e += 256) {
-            const int t = e / (BN * (FC_BK / 8));
-            const int rem = e % (BN * (FC_BK / 8));
-            const int n = rem / (FC_BK / 8);
-            const int c8 = (rem % (FC_BK / 8)) * 8;
-            uint4v v = {0, 0, 0, 0};
-            const int gn = n0 + n;
-            const int k = k0 + c8;
-            const int tap = AT ? t : (ty0 * KW + t);
-            if (gn < N && k + 8 <= Cin)
-                v = *(const uint4v*)(
-                    wp + ((size_t)tap * N + gn) * Cin + k);
-            else if (gn < N) {
-                __hip_bfloat16 tmp[8];
-                for (int u = 0; u < 8; ++u)
-                    tmp[u] = (k + u < Cin)
-                        ? wp[((size_t)tap * N + gn) * Cin + k + u]
-                        : (__hip_bfloat16)0.f;
-                v = *(const uint4v*)tmp;
+        // ---- weight tiles [NBT][BN][FC_BK]. Fast path for fully
+        // interior tiles (block-uniform: no per-chunk guards — they were
+        // ~9 guarded chunks/thread/step in the AT 3x3 kernel)
+        if (n0 + BN <= N && k0 + FC_BK <= Cin) {
+            for (int e = tid; e < NBT * BN * (FC_BK / 8); e += 256) {
+                const int t = e / (BN * (FC_BK / 8));
+                const int rem = e % (BN * (FC_BK / 8));
+                const int n = rem / (FC_BK / 8);
+                const int c8 = (rem % (FC_BK / 8)) * 8;
+                const int tap = AT ? t : (ty0 * KW + t);
+                const uint4v v = *(const uint4v*)(
+                    wp + ((size_t)tap * N + n0 + n) * Cin + k0 + c8);
+                *(uint4v*)(sBbase + t * BBYTES + fswz(n, c8 * 2)) = v;
             }
-            *(uint4v*)(sBbase + t * BBYTES + fswz(n, c8 * 2)) = v;
+        } else {
+            for (int e = tid; e < NBT * BN * (FC_BK / 8); e += 256) {
+                const int t = e / (BN * (FC_BK / 8));
+                const int rem = e % (BN * (FC_BK / 8));
+                const int n = rem / (FC_BK / 8);
+                const int c8 = (rem % (FC_BK / 8)) * 8;
+                uint4v v = {0, 0, 0, 0};
+                const int gn = n0 + n;
+                const int k = k0 + c8;
+                const int tap = AT ? t : (ty0 * KW + t);
+                if (gn < N && k + 8 <= Cin)
+                    v = *(const uint4v*)(
+                        wp + ((size_t)tap * N + gn) * Cin + k);
+                else if (gn < N) {
+                    __hip_bfloat16 tmp[8];
+                    for (int u = 0; u < 8; ++u)
+                        tmp[u] = (k + u < Cin)
+                            ? wp[((size_t)tap * N + gn) * Cin + k + u]
+                            : (__hip_bfloat16)0.f;
+                    v = *(const uint4v*)tmp;
+                }
+                *(uint4v*)(sBbase + t * BBYTES + fswz(n, c8 * 2)) = v;
+            }
         }
     };
 
