@@ -146,6 +146,34 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
         // ---- A slabs: NSLAB (= rows x m-tiles x parity) x AW x FC_BK ch
         const int Hi = (S == 2) ? 2 * H : H;     // input dims (even for S2)
         const int Wi = (S == 2) ? 2 * W : W;
+        // interior fast path (guide trap 4c: per-element guarded loads
+        // serialize — hipcc branches around each load and waits vmcnt(0)):
+        // all rows/cols in bounds AND this k-chunk entirely in one tensor.
+        const bool a_one = (k0 + FC_BK <= C1) || (k0 >= C1);
+        const bool a_rows =
+            (S == 1) && (y - KH / 2 + (AT ? 0 : ty0) >= 0) &&
+            (y + (AT ? KH - 1 : ty0) - KH / 2 < H);
+        const bool a_cols = (x0 - KW / 2 + (MT - 1) * 0 >= 0) &&
+                            (x0 + MT * BM - 1 + (KW - 1) / 2 < W);
+        if (a_one && a_rows && a_cols && S == 1) {
+            const bool use1 = k0 + FC_BK <= C1;
+            const __hip_bfloat16* src = use1 ? in1 : in2;
+            const int cs = use1 ? in1_stride : C2;
+            const int co = use1 ? (in1_off + k0) : (k0 - C1);
+            for (int e = tid; e < NSLAB * AW * (FC_BK / 8); e += 256) {
+                const int sl = e / (AW * (FC_BK / 8));
+                const int rem0 = e % (AW * (FC_BK / 8));
+                const int ar = rem0 / (FC_BK / 8);
+                const int c8 = (rem0 % (FC_BK / 8)) * 8;
+                const int mt = (sl / PAR) % MT;
+                const int rsl = sl / (PAR * MT);
+                const int row = y + (ty0 + rsl) - PBH;
+                const int x = x0 + mt * BM + ar - PBW;
+                const uint4v v = *(const uint4v*)(
+                    src + (((long long)b * H + row) * W + x) * cs + co + c8);
+                *(uint4v*)(sAbase + sl * ABYTES + fswz(ar, c8 * 2)) = v;
+            }
+        } else
         for (int e = tid; e < NSLAB * AW * (FC_BK / 8); e += 256) {
             const int sl = e / (AW * (FC_BK / 8));
             const int rem0 = e % (AW * (FC_BK / 8));
