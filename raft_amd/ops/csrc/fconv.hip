@@ -531,7 +531,34 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
     const long long groups = (ncells + PPB - 1) / PPB;
     for (long long g = blockIdx.x; g < groups; g += gridDim.x) {
         const long long cell0 = g * (long long)PPB;
-        // stage PPB windows (zero-padded SAME)
+        // stage PPB windows (zero-padded SAME). Interior fast path: all
+        // cells of the group in-bounds with full windows and no row wrap
+        // (guide trap 4c — per-element guarded loads serialize).
+        const int Hi = cstride * H;
+        const int Wi = cstride * W;
+        const int pb = (cstride == 2) ? (kh - 2) / 2 : kh / 2;
+        const int pbw = (cstride == 2) ? (kw - 2) / 2 : kw / 2;
+        const int x_f = (int)(cell0 % W);
+        const int y_f = (int)((cell0 / W) % H);
+        const bool interior =
+            (cell0 + PPB <= ncells) && (x_f + PPB <= W) &&
+            (cstride * y_f - pb >= 0) &&
+            (cstride * y_f + kh - 1 - pb < Hi) &&
+            (cstride * x_f - pbw >= 0) &&
+            (cstride * (x_f + PPB - 1) + kw - 1 - pbw < Wi);
+        if (interior) {
+            const int b0 = (int)(cell0 / ((long long)W * H));
+            for (int e = tid; e < PPB * taps * C; e += 256) {
+                const int p = e / (taps * C);
+                const int rem = e % (taps * C);
+                const int t = rem / C;
+                const int c = rem % C;
+                const int yy = cstride * y_f + t / kw - pb;
+                const int xx = cstride * (x_f + p) + t % kw - pbw;
+                sin[e] = in[(((long long)b0 * Hi + yy) * Wi + xx)
+                            * in_stride + in_off + c];
+            }
+        } else
         for (int e = tid; e < PPB * taps * C; e += 256) {
             const int p = e / (taps * C);
             const int rem = e % (taps * C);
@@ -543,12 +570,6 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
                 const int x = (int)(cell % W);
                 const int y = (int)((cell / W) % H);
                 const int b = (int)(cell / ((long long)W * H));
-                // cstride == 2: TF-SAME stride-2 (pad begin (K-2)/2,
-                // even input dims = 2H x 2W); else stride-1 SAME
-                const int Hi = cstride * H;
-                const int Wi = cstride * W;
-                const int pb = (cstride == 2) ? (kh - 2) / 2 : kh / 2;
-                const int pbw = (cstride == 2) ? (kw - 2) / 2 : kw / 2;
                 const int yy = cstride * y + t / kw - pb;
                 const int xx = cstride * x + t % kw - pbw;
                 if (yy >= 0 && yy < Hi && xx >= 0 && xx < Wi)
