@@ -20,11 +20,14 @@
 //     (n_off / out_cstride), which is how the per-iteration GRU input
 //     buffer x = [ctx | motion | flow] is assembled without copies.
 //
-// Geometry: 256 threads = 4 waves (2m x 2n), tile BM=64 positions of one
-// output row x BN=128 channels; wave tile 32x64 = 2x4 fragments of
-// mfma_f32_16x16x32_bf16. K-loop: for each kernel row ty and BK=32 channel
-// step, stage the (BM + kw - 1)-wide input slab and all kw weight tiles,
-// one barrier pair, then kw shifted-LDS-read MFMA groups.
+// Geometry: 256 threads = 4 waves (2m x 2n); tile 64x128 (big) or 32x64
+// (small — batch-1 grid saturation), selected by grid size. K-loop: per
+// (kernel row, BK=32 channel step) stage the halo'd input slab(s) and
+// weight tiles (AT variant: all taps per step), one barrier pair, then
+// shifted-LDS-read MFMA groups; stride-2 stages even/odd input columns as
+// parity slabs. Interior tiles take unguarded staging fast paths (per-
+// element guarded loads serialize — guide §5 trap 4c). Every structural
+// choice here is A/B-measured: see tools/bench_fconv.py and profiles/.
 
 #include "common.h"
 #include <hip/hip_bf16.h>
