@@ -171,3 +171,29 @@ class Prefetcher:
             d2 = im2.pin_memory().to(self.device, self.dtype,
                                      non_blocking=True)
         return d1, d2
+
+
+def run_mixed_batch(engine: InferenceEngine, pairs, iters=None):
+    """Mixed-batch inference with per-sample dynamic H x W (BASELINE
+    config 5): groups same-shape pairs into batches (one kernel-efficient
+    call per shape group), preserves input order in the returned list.
+
+    pairs: sequence of (im1, im2) tensors [3,H,W] or [1,3,H,W].
+    """
+    norm = []
+    for im1, im2 in pairs:
+        if im1.dim() == 3:
+            im1 = im1.unsqueeze(0)
+            im2 = im2.unsqueeze(0)
+        norm.append((im1, im2))
+    groups = {}
+    for idx, (im1, im2) in enumerate(norm):
+        groups.setdefault(tuple(im1.shape[-2:]), []).append(idx)
+    out = [None] * len(norm)
+    for shape, idxs in groups.items():
+        b1 = torch.cat([norm[i][0] for i in idxs], dim=0)
+        b2 = torch.cat([norm[i][1] for i in idxs], dim=0)
+        flows = engine(b1, b2, iters=iters)
+        for j, i in enumerate(idxs):
+            out[i] = flows[j:j + 1]
+    return out
