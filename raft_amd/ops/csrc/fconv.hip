@@ -452,7 +452,11 @@ extern "C" void launch_fconv_nhwc_bf16(
     // per-block weight staging but MEASURED worse nearly everywhere
     // (tools/bench_fconv.py round14: grid saturation dominates at these
     // batch-1 sizes) — kept selectable for larger-batch shapes.
-    const int mt = (mtiles < 0) ? 1 : mtiles;
+    // mtiles == -2: force the big (64x128) tile regardless of grid size
+    // (structure-efficiency probes).
+    const int mt = (mtiles < -1 || mtiles == 0) ? 1
+                   : (mtiles < 0 ? 1 : mtiles);
+    const bool force_big = (mtiles == -2);
 #define FC_LAUNCH(KH, KW, MI, NJ, AT, MTv, BMv, BNv)                         \
     {                                                                        \
         dim3 grid(cdiv(N, BNv), H * cdiv(W, (BMv) * (MTv)), B);              \
@@ -488,7 +492,7 @@ extern "C" void launch_fconv_nhwc_bf16(
     }
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
-        if (big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)                  \
+        if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
         if (KH > 1 && at) {                                                  \
             if (mt >= 4) FC_LAUNCH(KH, KW, 1, 2, true, 4, 32, 64)            \
             if (mt == 2) FC_LAUNCH(KH, KW, 1, 2, true, 2, 32, 64)            \
