@@ -384,3 +384,42 @@ def test_fused_model_shape_sweep(dev, h, w):
         out = m(x1, x2, iters=3)
     err = (out.float() - ref.float()).abs().max().item()
     assert err < 0.05, (h, w, err)
+
+
+def test_fused_determinism_race_screen(dev):
+    """Same input twice must be bit-identical — a practical race screen
+    for the kernel set (atomics are absent from the inference path)."""
+    from raft_amd import RAFT, RaftConfig
+    m = RAFT(RaftConfig(small=False)).to(dev).to(torch.bfloat16).eval()
+    for h, w in ((64, 96), (436, 1024)):
+        x1 = torch.rand(1, 3, h, w, device=dev, dtype=torch.bfloat16)
+        x2 = torch.rand(1, 3, h, w, device=dev, dtype=torch.bfloat16)
+        with torch.no_grad():
+            a = m(x1, x2, iters=4).clone()
+            b = m(x1, x2, iters=4)
+        assert torch.equal(a, b), (h, w)
+
+
+def test_fused_model_full_res_headline(dev):
+    """Fused vs eager-bf16 at the exact headline resolution (436x1024 ->
+    pad8 440x1024) — tile-boundary behavior at the benchmarked shape."""
+    import os as _os
+
+    from raft_amd import RAFT, RaftConfig
+    from raft_amd.engine.inference import pad8
+    torch.manual_seed(3)
+    m = RAFT(RaftConfig(small=False)).to(dev).to(torch.bfloat16).eval()
+    x1, _ = pad8(torch.rand(1, 3, 436, 1024, device=dev,
+                            dtype=torch.bfloat16))
+    x2, _ = pad8(torch.rand(1, 3, 436, 1024, device=dev,
+                            dtype=torch.bfloat16))
+    with torch.no_grad():
+        _os.environ["RAFT_AMD_NO_FUSE"] = "1"
+        try:
+            ref = m(x1, x2, iters=3)
+        finally:
+            _os.environ.pop("RAFT_AMD_NO_FUSE")
+        out = m(x1, x2, iters=3)
+    err = (out.float() - ref.float()).abs()
+    assert err.max().item() < 0.25, err.max().item()
+    assert err.mean().item() < 0.01, err.mean().item()
