@@ -50,6 +50,7 @@ void launch_fconv_nhwc_bf16(const void*, int, int, int, const void*, int,
                             hipStream_t);
 void launch_inorm_stats(const void*, float*, float*, float*, int, int,
                         int, float, hipStream_t);
+int inorm_stats_partitions(int);
 void launch_inorm_apply(const void*, const float*, const float*,
                         const void*, void*, int, int, int, int,
                         hipStream_t);
@@ -442,7 +443,8 @@ std::vector<at::Tensor> inorm_stats(at::Tensor in) {
     CHECK_DEV(in); CHECK_CONT(in);
     const int B = in.size(0), H = in.size(1), W = in.size(2);
     const int C = in.size(3);
-    auto acc = at::zeros({B, C, 2}, in.options().dtype(at::kFloat));
+    const int S = inorm_stats_partitions(H * W);
+    auto acc = at::empty({B, C, S, 2}, in.options().dtype(at::kFloat));
     auto mean = at::empty({B, C}, in.options().dtype(at::kFloat));
     auto rstd = at::empty({B, C}, in.options().dtype(at::kFloat));
     launch_inorm_stats(in.data_ptr(), acc.data_ptr<float>(),

@@ -9,12 +9,13 @@
 #include <hip/hip_bf16.h>
 
 // stats, pass 1: split reduction — grid (B, ceil(C/64), S position
-// partitions); each block reduces its partition into LDS then one
-// atomicAdd per (c, moment) into the zero-initialized accumulator.
+// partitions); each block writes its partition's moments to its OWN slab
+// slot (no atomics: float atomicAdd order made the whole fused path
+// nondeterministic — caught by the bit-determinism race screen).
 // (A (B, C/64) grid was 2 blocks on a 256-CU chip: 3.2 ms/call.)
 extern "C" __global__ __launch_bounds__(256) void inorm_stats_part_k(
     const __hip_bfloat16* __restrict__ in,   // [B, H*W, C]
-    float* __restrict__ acc,                 // [B, C, 2] zero-initialized
+    float* __restrict__ acc,                 // [B, C, S, 2] slabs
     int HW, int C) {
     __shared__ float red[2][4][64];
     const int b = blockIdx.x;
@@ -40,19 +41,27 @@ extern "C" __global__ __launch_bounds__(256) void inorm_stats_part_k(
             ts += red[0][i][threadIdx.x & 63];
             ts2 += red[1][i][threadIdx.x & 63];
         }
-        atomicAdd(&acc[((size_t)b * C + c) * 2], ts);
-        atomicAdd(&acc[((size_t)b * C + c) * 2 + 1], ts2);
+        const size_t slot = (((size_t)b * C + c) * S + blockIdx.z) * 2;
+        acc[slot] = ts;
+        acc[slot + 1] = ts2;
     }
 }
 
-// stats, pass 2: finalize mean/rstd from the accumulated moments
+// stats, pass 2: finalize mean/rstd — fixed-order sum over the S slabs
+// (deterministic by construction)
 extern "C" __global__ void inorm_stats_fin_k(
     const float* __restrict__ acc, float* __restrict__ mean,
-    float* __restrict__ rstd, int HW, long long BC, float eps) {
+    float* __restrict__ rstd, int HW, int S, long long BC, float eps) {
     const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= BC) return;
-    const float m = acc[i * 2] / HW;
-    const float var = fmaxf(acc[i * 2 + 1] / HW - m * m, 0.0f);
+    float ts = 0.f, ts2 = 0.f;
+    const float* slab = acc + (size_t)i * S * 2;
+    for (int z = 0; z < S; ++z) {
+        ts += slab[z * 2];
+        ts2 += slab[z * 2 + 1];
+    }
+    const float m = ts / HW;
+    const float var = fmaxf(ts2 / HW - m * m, 0.0f);
     mean[i] = m;
     rstd[i] = rsqrtf(var + eps);
 }
@@ -79,17 +88,21 @@ extern "C" __global__ void inorm_apply_k(
     }
 }
 
+extern "C" int inorm_stats_partitions(int HW) {
+    return (int)min((long long)cdiv(HW, 1024), (long long)128);
+}
+
 extern "C" void launch_inorm_stats(const void* in, float* acc, float* mean,
                                    float* rstd, int B, int HW, int C,
                                    float eps, hipStream_t s) {
-    const int S = (int)min((long long)cdiv(HW, 1024), (long long)128);
+    const int S = inorm_stats_partitions(HW);
     dim3 grid(B, cdiv(C, 64), S);
     hipLaunchKernelGGL(inorm_stats_part_k, grid, dim3(256), 0, s,
                        (const __hip_bfloat16*)in, acc, HW, C);
     const long long BC = (long long)B * C;
     hipLaunchKernelGGL(inorm_stats_fin_k,
                        dim3((unsigned)((BC + 255) / 256)), dim3(256), 0, s,
-                       acc, mean, rstd, HW, BC, eps);
+                       acc, mean, rstd, HW, S, BC, eps);
 }
 
 extern "C" void launch_inorm_apply(const void* in, const float* mean,
