@@ -17,12 +17,16 @@ _PNG_SIG = b"\x89PNG\r\n\x1a\n"
 
 
 def read_png(path: str) -> np.ndarray:
-    """Read a PNG into an HxWx3 uint8 **BGR** array (the channel order the
-    model weights expect — networks/RAFT.py:13)."""
+    """Read a PNG file into an HxWx3 uint8 **BGR** array (the channel
+    order the model weights expect — networks/RAFT.py:13)."""
     with open(path, "rb") as f:
-        data = f.read()
+        return decode_png(f.read())
+
+
+def decode_png(data: bytes) -> np.ndarray:
+    """Decode in-memory PNG bytes to HxWx3 uint8 BGR."""
     if data[:8] != _PNG_SIG:
-        raise ValueError(f"{path}: not a PNG")
+        raise ValueError("not a PNG")
     pos = 8
     width = height = bit_depth = color_type = None
     idat = bytearray()
@@ -35,9 +39,9 @@ def read_png(path: str) -> np.ndarray:
             width, height, bit_depth, color_type, _, _, interlace = \
                 struct.unpack(">IIBBBBB", chunk)
             if bit_depth != 8:
-                raise ValueError(f"{path}: only 8-bit PNGs supported")
+                raise ValueError(f"only 8-bit PNGs supported")
             if interlace:
-                raise ValueError(f"{path}: interlaced PNGs not supported")
+                raise ValueError(f"interlaced PNGs not supported")
         elif ctype == b"PLTE":
             palette = np.frombuffer(chunk, np.uint8).reshape(-1, 3)
         elif ctype == b"IDAT":
@@ -49,7 +53,7 @@ def read_png(path: str) -> np.ndarray:
     stride = width * channels
     expected = height * (stride + 1)
     if len(raw) != expected:
-        raise ValueError(f"{path}: bad IDAT size {len(raw)} != {expected}")
+        raise ValueError(f"bad IDAT size {len(raw)} != {expected}")
     raw = np.frombuffer(raw, np.uint8).reshape(height, stride + 1)
     filters = raw[:, 0]
     img = _unfilter(raw[:, 1:].astype(np.int32), filters, channels)
@@ -112,7 +116,13 @@ def _unfilter(rows: np.ndarray, filters: np.ndarray, bpp: int) -> np.ndarray:
 
 
 def write_png(path: str, img: np.ndarray) -> None:
-    """Write an HxWx3 uint8 **BGR** array as a PNG (RGB8, filter 0)."""
+    """Write an HxWx3 uint8 **BGR** array as a PNG file (RGB8, filter 0)."""
+    with open(path, "wb") as f:
+        f.write(encode_png(img))
+
+
+def encode_png(img: np.ndarray) -> bytes:
+    """Encode an HxWx3 uint8 BGR array to PNG bytes (RGB8, filter 0)."""
     if img.dtype != np.uint8:
         img = np.clip(img, 0, 255).astype(np.uint8)
     if img.ndim == 2:
@@ -127,8 +137,5 @@ def write_png(path: str, img: np.ndarray) -> None:
             struct.pack(">I", crc)
 
     ihdr = struct.pack(">IIBBBBB", w, h, 8, 2, 0, 0, 0)
-    with open(path, "wb") as f:
-        f.write(_PNG_SIG)
-        f.write(chunk(b"IHDR", ihdr))
-        f.write(chunk(b"IDAT", zlib.compress(raw, 6)))
-        f.write(chunk(b"IEND", b""))
+    return (_PNG_SIG + chunk(b"IHDR", ihdr) +
+            chunk(b"IDAT", zlib.compress(raw, 6)) + chunk(b"IEND", b""))

@@ -29,14 +29,9 @@ except ImportError:  # pragma: no cover
 
 
 def _decode_png_bytes(data: bytes) -> torch.Tensor:
-    """PNG bytes -> [1,3,H,W] float BGR in [0,1] (via the in-repo codec)."""
-    import tempfile
-
-    from raft_amd.data.imageio import read_png
-    with tempfile.NamedTemporaryFile(suffix=".png") as f:
-        f.write(data)
-        f.flush()
-        img = read_png(f.name)
+    """PNG bytes -> [1,3,H,W] float BGR in [0,1] (in-memory codec)."""
+    from raft_amd.data.imageio import decode_png
+    img = decode_png(data)
     t = torch.from_numpy(img.astype(np.float32) / 255.0)
     return t.permute(2, 0, 1).unsqueeze(0)
 
@@ -90,13 +85,9 @@ def create_app(model=None, iters: Optional[int] = None,
             flow_np = out[0].float().permute(1, 2, 0).cpu().numpy()
             if fmt == "color":
                 from raft_amd.utils.flow_viz import flow_to_color
-                from raft_amd.data.imageio import write_png
-                import tempfile
+                from raft_amd.data.imageio import encode_png
                 color = flow_to_color(flow_np, convert_to_bgr=True)
-                with tempfile.NamedTemporaryFile(suffix=".png") as f:
-                    write_png(f.name, color)
-                    f.seek(0)
-                    payload = open(f.name, "rb").read()
+                payload = encode_png(color)
                 media = "image/png"
             else:
                 payload = _flo_bytes(flow_np)
