@@ -139,20 +139,32 @@ def mode_val(args, device):
     from raft_amd.data.synthetic import synthetic_pair
     from raft_amd.engine.trainer import epe
     from raft_amd.engine.inference import InferenceEngine
+    from raft_amd.parallel.ddp import init_distributed
+    import torch.distributed as dist
 
+    rank = init_distributed()  # DP eval: shard seeds, all-reduce the metric
+    world = dist.get_world_size() if dist.is_initialized() else 1
     model = _build_model(args, device)
     engine = InferenceEngine(
         model, iters=args.iters,
         dtype=torch.bfloat16 if args.dtype == "bf16" else torch.float32,
         use_graph=not args.no_graph)
     epes = []
-    for seed in range(8):
+    for seed in range(8)[rank::world]:
         im1, im2, gt = synthetic_pair(args.batch, 288, 512, seed=seed)
         flow = engine(im1, im2)
         epes.append(float(epe(flow.float().cpu(), gt)))
-    result = {"epe_mean": float(np.mean(epes)), "epe_per_batch": epes,
-              "data": "synthetic-warp 288x512", "iters": args.iters}
-    print(json.dumps(result))
+    if world > 1:
+        t = torch.tensor([float(np.sum(epes)), float(len(epes))],
+                         dtype=torch.float64, device=device)
+        dist.all_reduce(t)   # per-rank metric reduction
+        mean = float((t[0] / t[1]).item())
+    else:
+        mean = float(np.mean(epes))
+    if rank == 0:
+        result = {"epe_mean": mean, "epe_per_batch": epes, "world": world,
+                  "data": "synthetic-warp 288x512", "iters": args.iters}
+        print(json.dumps(result))
 
 
 def mode_train(args, device):

@@ -71,17 +71,91 @@ class PairDataflow:
 
 
 # ---------------------------------------------------------------- training aug
+_DCT8: Optional[torch.Tensor] = None
+
+# standard JPEG luminance quantization table (ITU-T T.81 Annex K.1)
+_JPEG_Q = torch.tensor(
+    [[16, 11, 10, 16, 24, 40, 51, 61],
+     [12, 12, 14, 19, 26, 58, 60, 55],
+     [14, 13, 16, 24, 40, 57, 69, 56],
+     [14, 17, 22, 29, 51, 87, 80, 62],
+     [18, 22, 37, 56, 68, 109, 103, 77],
+     [24, 35, 55, 64, 81, 104, 113, 92],
+     [49, 64, 78, 87, 103, 121, 120, 101],
+     [72, 92, 95, 98, 112, 100, 103, 99]], dtype=torch.float32)
+
+
+def _dct8() -> torch.Tensor:
+    global _DCT8
+    if _DCT8 is None:
+        import math
+        k = torch.arange(8, dtype=torch.float32)
+        D = torch.cos((2 * k[None, :] + 1) * k[:, None] * math.pi / 16) * 0.5
+        D[0] *= 1.0 / math.sqrt(2.0)
+        _DCT8 = D
+    return _DCT8
+
+
+def gaussian_blur(img: torch.Tensor, sigma: float) -> torch.Tensor:
+    """Separable Gaussian blur of a [...,C,H,W] image in [0,1]."""
+    r = max(1, int(3.0 * sigma + 0.5))
+    x = torch.arange(-r, r + 1, dtype=torch.float32)
+    k = torch.exp(-x * x / (2.0 * sigma * sigma))
+    k = k / k.sum()
+    v = img if img.dim() == 4 else img[None]
+    C = v.shape[1]
+    kw = k.view(1, 1, 1, -1).expand(C, 1, 1, 2 * r + 1)
+    kh = k.view(1, 1, -1, 1).expand(C, 1, 2 * r + 1, 1)
+    v = F.conv2d(F.pad(v, (r, r, 0, 0), mode="replicate"), kw, groups=C)
+    v = F.conv2d(F.pad(v, (0, 0, r, r), mode="replicate"), kh, groups=C)
+    return v if img.dim() == 4 else v[0]
+
+
+def jpeg_noise(img: torch.Tensor, quality: float) -> torch.Tensor:
+    """JPEG compression artifacts for a [...,C,H,W] image in [0,1]:
+    8x8 blockwise DCT-II, quantization by the standard luminance table
+    scaled to `quality` (1..100), inverse DCT.  Per-channel (no chroma
+    subsampling) — the artifact structure the reference's cv2
+    JpegNoise augmentor injects (test_dataflow.py:29-34), without a
+    JPEG codec in the image."""
+    v = img if img.dim() == 4 else img[None]
+    B, C, H, W = v.shape
+    ph, pw = (-H) % 8, (-W) % 8
+    x = F.pad(v * 255.0 - 128.0, (0, pw, 0, ph), mode="replicate")
+    Hp, Wp = x.shape[-2:]
+    blk = x.reshape(B * C, Hp // 8, 8, Wp // 8, 8).permute(0, 1, 3, 2, 4)
+    D = _dct8()
+    coef = D @ blk @ D.T
+    scale = 5000.0 / quality if quality < 50 else 200.0 - 2.0 * quality
+    q = torch.clamp(_JPEG_Q * scale / 100.0, min=1.0)
+    coef = torch.round(coef / q) * q
+    blk = D.T @ coef @ D
+    x = blk.permute(0, 1, 3, 2, 4).reshape(B, C, Hp, Wp)[:, :, :H, :W]
+    out = ((x + 128.0) / 255.0).clamp(0, 1)
+    return out if img.dim() == 4 else out[0]
+
+
 def augment_pair(im1: torch.Tensor, im2: torch.Tensor, flow: torch.Tensor,
                  g: torch.Generator,
                  crop: Optional[Tuple[int, int]] = None):
     """Shared-parameter photometric + geometric augmentation of a pair
-    (reference intent: test_dataflow.py:20-41 — contrast/gamma shared across
-    the pair, h-flip with frame swap, random crop)."""
+    (reference intent: test_dataflow.py:20-41 — contrast/gamma/blur/jpeg
+    noise shared across the pair, h-flip with frame swap, random crop)."""
     # photometric: shared contrast & gamma
     c = 0.8 + 0.4 * torch.rand((), generator=g).item()
     gamma = 0.8 + 0.4 * torch.rand((), generator=g).item()
     im1 = (im1 * c).clamp(0, 1) ** gamma
     im2 = (im2 * c).clamp(0, 1) ** gamma
+    # shared-sigma Gaussian blur (reference: GaussianBlur, shared params)
+    if torch.rand((), generator=g).item() < 0.3:
+        sigma = 0.5 + 1.0 * torch.rand((), generator=g).item()
+        im1 = gaussian_blur(im1, sigma)
+        im2 = gaussian_blur(im2, sigma)
+    # shared-quality JPEG compression noise (reference: JpegNoise)
+    if torch.rand((), generator=g).item() < 0.3:
+        quality = 40.0 + 50.0 * torch.rand((), generator=g).item()
+        im1 = jpeg_noise(im1, quality)
+        im2 = jpeg_noise(im2, quality)
     # horizontal flip (x component negates; the reference's frame-order swap
     # variant only applies without ground truth — it would invalidate flow)
     if torch.rand((), generator=g).item() < 0.5:

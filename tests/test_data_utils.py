@@ -158,3 +158,59 @@ def test_run_mixed_batch_groups_and_order():
     # batched group of shape (40,56) contains samples 0 and 2; batching
     # is numerically equivalent for this model (no cross-sample ops)
     assert torch.allclose(outs[0], solo, atol=1e-4)
+
+
+def test_gaussian_blur_smooths_and_preserves_shape():
+    from raft_amd.data.dataflow import gaussian_blur
+    g = torch.Generator().manual_seed(0)
+    img = torch.rand(3, 37, 53, generator=g)
+    out = gaussian_blur(img, sigma=1.2)
+    assert out.shape == img.shape
+    # blur reduces high-frequency energy (neighbor differences)
+    def hf(x):
+        return (x[..., 1:] - x[..., :-1]).abs().mean()
+    assert hf(out) < hf(img) * 0.6
+    # flat image is (near) invariant — replicate padding, normalized kernel
+    flat = torch.full((3, 16, 16), 0.5)
+    assert torch.allclose(gaussian_blur(flat, 1.0), flat, atol=1e-6)
+    # batched input works
+    assert gaussian_blur(img[None], 1.0).shape == (1, 3, 37, 53)
+
+
+def test_jpeg_noise_quality_monotone_and_shape():
+    from raft_amd.data.dataflow import jpeg_noise
+    g = torch.Generator().manual_seed(1)
+    img = torch.rand(3, 29, 43, generator=g)  # non-multiple-of-8 on purpose
+    hi = jpeg_noise(img, 95.0)
+    lo = jpeg_noise(img, 20.0)
+    assert hi.shape == img.shape and lo.shape == img.shape
+    assert hi.min() >= 0 and hi.max() <= 1
+    err_hi = (hi - img).abs().mean()
+    err_lo = (lo - img).abs().mean()
+    assert err_hi < err_lo, "lower quality must distort more"
+    assert err_hi < 0.05
+
+
+def test_augment_pair_shared_params_and_flow_consistency():
+    from raft_amd.data.dataflow import augment_pair
+    from raft_amd.engine.trainer import epe
+    # the same seed must give identical augmentation twice (shared RNG)
+    im1 = torch.rand(1, 3, 64, 96)
+    im2 = torch.rand(1, 3, 64, 96)
+    flow = torch.randn(1, 2, 64, 96)
+    outs = []
+    for _ in range(2):
+        g = torch.Generator().manual_seed(7)
+        outs.append(augment_pair(im1.clone(), im2.clone(), flow.clone(), g,
+                                 crop=(48, 64)))
+    for a, b in zip(outs[0], outs[1]):
+        assert torch.equal(a, b)
+    a1, a2, af = outs[0]
+    assert a1.shape == (1, 3, 48, 64) and af.shape == (1, 2, 48, 64)
+    # photometric params are shared: warping consistency of a constant pair
+    # (augmenting identical frames keeps them identical)
+    same = torch.rand(1, 3, 64, 96)
+    for seed in range(6):   # cover blur/jpeg probability branches
+        g = torch.Generator().manual_seed(seed)
+        b1, b2, _ = augment_pair(same.clone(), same.clone(), flow.clone(), g)
+        assert torch.equal(b1, b2)
