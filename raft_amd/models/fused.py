@@ -104,7 +104,7 @@ class FusedBasicUpdate:
         self.m2 = _FC(ub.mask[2], scale=0.25)        # fold the 0.25 scale
         self.ctx_dim = ctx_dim
 
-    def __call__(self, hip, net, x_buf, corr_pad, coords1):
+    def __call__(self, hip, net, x_buf, corr_pad, coords1, final=True):
         # motion encoder (model_utils.py:110-119). The flow channels of
         # x_buf were already written by the lookup kernel (fused flow out).
         ctx = self.ctx_dim
@@ -117,7 +117,17 @@ class FusedBasicUpdate:
         # SepConvGRU (model_utils.py:138-156)
         net = self.gru1(hip, net, x_buf)
         net = self.gru2(hip, net, x_buf)
-        # heads: one merged 3x3 conv; mask from a strided slice; the
+        if not final:
+            # intermediate iterations: the mask head is dead in test mode
+            # (only the FINAL flow is upsampled — exactly the pruning TF
+            # performs on the reference's fetched graph), so run the
+            # flow-head 3x3 alone instead of the merged N=512 conv
+            h1 = self.fh1(hip, net)
+            coords_new = hip.fconv_dflow_coords(h1, self.fh2.wp,
+                                                self.fh2.bias, coords1, 3, 3)
+            return net, None, coords_new
+        # final iteration — heads: one merged 3x3 conv (flow_head.conv1 +
+        # mask[0] stacked along N); mask from a strided slice; the
         # delta-flow final conv also applies coords1 += dflow in-kernel
         hbuf = hip.fconv_plain(net, None, self.heads_w, self.heads_b, 3, 3,
                                ACT_RELU, None, 0, 0, 0, -1, -1, 1, None)
@@ -142,7 +152,7 @@ class FusedSmallUpdate:
         self.ctx_dim = ctx_dim
         # x = [inp(ctx) | motion(80) | flow(2)]; motion encoder out = 80
 
-    def __call__(self, hip, net, x_buf, corr_pad, coords1):
+    def __call__(self, hip, net, x_buf, corr_pad, coords1, final=True):
         ctx = self.ctx_dim
         cor = self.c1(hip, corr_pad)
         flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
@@ -327,14 +337,15 @@ class FusedRaft:
                                device=net.device, dtype=torch.bfloat16)
         flow_off = self.x_dim - 2      # flow channels of the GRU input
         mask = None
-        for _ in range(iters):
+        for it in range(iters):
             # lookup writes the taps AND flow = coords1 - grid directly
             # into the GRU input buffer's flow slice (zero glue kernels)
             corr_pad = hip.corr_lookup_nhwc(list(levels), coords1,
                                             cfg.corr_radius, self.corr_cpad,
                                             True, corr_buf, x_buf, flow_off)
             net, mask, coords1 = self.update(hip, net, x_buf, corr_pad,
-                                             coords1)
+                                             coords1,
+                                             final=(it == iters - 1))
 
         flow = coords1 - coords0                         # [B,H,W,2] fp32
         if cfg.small:
