@@ -501,7 +501,9 @@ extern "C" void launch_fconv_nhwc_bf16(
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
         if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
-        if (tile11) {                                                        \
+        if (tile11 && !(KH == 5 && KW == 1)) {                               \
+            /* 5x1 excluded: vertical taps share no staged rows, so the  */ \
+            /* extra workgroups just duplicate A slabs (39.1 vs 37.7 us) */ \
             if (KH > 1 && at) FC_LAUNCH(KH, KW, 1, 1, true, 1, 32, 32)       \
             FC_LAUNCH(KH, KW, 1, 1, false, 1, 32, 32)                        \
         }                                                                    \
@@ -639,6 +641,29 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallc_tile_k(
     const int x0 = (blockIdx.x % xt) * 16;
     const int n0 = blockIdx.y * 64;
     const int b = blockIdx.z;
+    if ((N & 63) == 0) {
+        // vectorized: the 64xC weight slice of each tap is contiguous in
+        // wp and 16-byte aligned (N, n0 multiples of 64) — 8 bf16 per
+        // load. (The scalar loop below was ~24 serial 2-byte rounds: the
+        // dominant cost of the whole kernel at C=2.)
+        const int per_tap = 64 * C / 8;
+        for (int e8 = tid; e8 < taps * per_tap; e8 += 256) {
+            const int t = e8 / per_tap;
+            const int j = (e8 % per_tap) * 8;
+            const uint4v v = *(const uint4v*)(
+                wp + ((size_t)t * N + n0) * C + j);
+            const unsigned int w4[4] = {v.x, v.y, v.z, v.w};
+            float* dst = sw + t * 64 * C + j;
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+                union { unsigned int i; float f; } lo, hi;
+                lo.i = (w4[q] & 0xffffu) << 16;
+                hi.i = (w4[q] >> 16) << 16;
+                dst[2 * q] = lo.f;
+                dst[2 * q + 1] = hi.f;
+            }
+        }
+    } else
     for (int e = tid; e < taps * 64 * C; e += 256) {
         const int t = e / (64 * C);
         const int rem = e % (64 * C);
