@@ -82,7 +82,11 @@ RAFT_DEV float factivate(float v, int act) {
 // consecutive-row (conflict-free swizzle) — input cols 2m+tx map to
 // parity (tx-PB)&1, row offset (tx-PB)>>1. Requires even input dims
 // (TF-SAME pad (K-2)/2 begin — Conv2dTF semantics for even inputs).
-template <int KH, int KW, int MI, int NJ, bool AT, int MT, int S>
+// TH: output rows per tile (2 = 2 rows x 16 cols instead of 1 x 32, AT
+// and MI=1 only): vertical taps then SHARE staged row slabs — a 5x1 conv
+// stages KH+1=6 slabs of 16 positions per 32 outputs instead of 5 slabs
+// of 32 (1.7x less A staging), a 3x3 stages 4x18 instead of 3x34.
+template <int KH, int KW, int MI, int NJ, bool AT, int MT, int S, int TH = 1>
 __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     const __hip_bfloat16* __restrict__ in1, int C1,
     int in1_stride, int in1_off,                      // strided slice of in1
@@ -102,10 +106,11 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
     constexpr int PBW = (S == 2) ? (KW - 2) / 2 : KW / 2;  // left pad
     constexpr int PBH = (S == 2) ? (KH - 2) / 2 : KH / 2;
     constexpr int RLO = (S == 2) ? (PBW + 1) / 2 : 0;
+    constexpr int BMX = BM / TH;         // x extent of the position tile
     constexpr int AW = (S == 2) ? (BM + RLO + (KW + 1) / 2 + 1)
-                                : (BM + KW - 1);   // slab rows (+halo)
+                                : (BMX + KW - 1);  // slab rows (+halo)
     constexpr int APAD = ((AW + 15) / 16) * 16;  // swizzle window rounding
-    constexpr int NSLAB = (AT ? KH : 1) * MT * PAR;
+    constexpr int NSLAB = ((AT ? KH : 1) + TH - 1) * MT * PAR;
     constexpr int NBT = AT ? TAPS : KW;
     constexpr int ABYTES = APAD * FC_ROWB;
     constexpr int BBYTES = BN * FC_ROWB;
@@ -116,9 +121,9 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
 
     const int Cin = C1 + C2;
     const int b = blockIdx.z;
-    const int tiles_per_row = (W + MT * BM - 1) / (MT * BM);
-    const int y = blockIdx.y / tiles_per_row;
-    const int x0 = (blockIdx.y % tiles_per_row) * (MT * BM);
+    const int tiles_per_row = (W + MT * BMX - 1) / (MT * BMX);
+    const int y = TH * (blockIdx.y / tiles_per_row);
+    const int x0 = (blockIdx.y % tiles_per_row) * (MT * BMX);
     const int n0 = blockIdx.x * BN;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -155,9 +160,9 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
         const bool a_one = (k0 + FC_BK <= C1) || (k0 >= C1);
         const bool a_rows =
             (S == 1) && (y - KH / 2 + (AT ? 0 : ty0) >= 0) &&
-            (y + (AT ? KH - 1 : ty0) - KH / 2 < H);
+            (y + (AT ? KH - 1 : ty0) + TH - 1 - KH / 2 < H);
         const bool a_cols = (x0 - KW / 2 + (MT - 1) * 0 >= 0) &&
-                            (x0 + MT * BM - 1 + (KW - 1) / 2 < W);
+                            (x0 + MT * BMX - 1 + (KW - 1) / 2 < W);
         if (a_one && a_rows && a_cols && S == 1) {
             const bool use1 = k0 + FC_BK <= C1;
             const __hip_bfloat16* src = use1 ? in1 : in2;
@@ -171,7 +176,7 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                 const int mt = (sl / PAR) % MT;
                 const int rsl = sl / (PAR * MT);
                 const int row = y + (ty0 + rsl) - PBH;
-                const int x = x0 + mt * BM + ar - PBW;
+                const int x = x0 + mt * BMX + ar - PBW;
                 const uint4v v = *(const uint4v*)(
                     src + (((long long)b * H + row) * W + x) * cs + co + c8);
                 *(uint4v*)(sAbase + sl * ABYTES + fswz(ar, c8 * 2)) = v;
@@ -190,8 +195,8 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                                      : (y + (ty0 + rsl) - PBH);
             const bool row_ok = (row >= 0 && row < Hi);
             const int x = (S == 2)
-                ? (2 * (x0 + mt * BM + ar - RLO) + par)
-                : (x0 + mt * BM + ar - PBW);
+                ? (2 * (x0 + mt * BMX + ar - RLO) + par)
+                : (x0 + mt * BMX + ar - PBW);
             uint4v v = {0, 0, 0, 0};
             if (row_ok && x >= 0 && x < Wi) {
                 const int k = k0 + c8;
@@ -280,13 +285,15 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                     const int d = tx - PBW;
                     const int par = (S == 2) ? (d & 1) : 0;
                     const int roff = (S == 2) ? ((d >> 1) + RLO) : (d + PBW);
+                    const int rslot = rsl + (TH == 2 ? (wm >> 4) : 0);
                     const char* sA = sAbase +
-                        ((rsl * MT + mt) * PAR + par) * ABYTES;
+                        ((rslot * MT + mt) * PAR + par) * ABYTES;
                     short8 af[MI], bf[NJ];
 #pragma unroll
                     for (int i = 0; i < MI; ++i)
                         af[i] = *(const short8*)(
-                            sA + fswz(wm + i * 16 + (lane & 15) + roff, cb));
+                            sA + fswz(((TH == 2) ? 0 : wm + i * 16)
+                                      + (lane & 15) + roff, cb));
 #pragma unroll
                     for (int j = 0; j < NJ; ++j)
                         bf[j] = *(const short8*)(
@@ -317,9 +324,11 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             for (int r = 0; r < 4; ++r) {
                 const int m = wm + i * 16 + (lane >> 4) * 4 + r;
                 const int n = n0 + wn + j * 16 + (lane & 15);
-                const int x = x0 + mt * BM + m;
-                if (x >= W || n >= N) continue;
-                const long long p = ((long long)b * H + y) * W + x;
+                const int yy2 = (TH == 2) ? y + (m >> 4) : y;
+                const int x = (TH == 2) ? (x0 + (m & 15))
+                                        : (x0 + mt * BM + m);
+                if (x >= W || n >= N || yy2 >= H) continue;
+                const long long p = ((long long)b * H + yy2) * W + x;
                 float v = acc[mt][i][j][r];
                 if (bias) v += bias[n];
                 if (mode == EP_PLAIN) {
@@ -498,9 +507,31 @@ extern "C" void launch_fconv_nhwc_bf16(
         const char* e = getenv("RAFT_AMD_TILE11");
         return e ? atoi(e) : 1;
     }();
+#define FC_LAUNCH_TH2(KH, KW, NJ, BNv)                                      \
+    {                                                                        \
+        dim3 grid(cdiv(N, BNv), ((H + 1) / 2) * cdiv(W, 16), B);             \
+        hipLaunchKernelGGL(                                                  \
+            (fconv_nhwc_bf16_k<KH, KW, 1, NJ, true, 1, 1, 2>), grid, blk,    \
+            0, s, FCONV_ARGS);                                               \
+        return;                                                              \
+    }
+    // 2-row tiles for the vertical-halo shapes (KH>1): measured 11.98 ->
+    // 11.77 ms/step on the headline config (3x3 24.5 -> 21.9 us, 5x1
+    // 37.7 -> 35.8). RAFT_AMD_TILE2D=0 disables, =2 selects the BN=64
+    // variant (measured worse: LDS-limited occupancy).
+    static const int tile2d = [] {
+        const char* e = getenv("RAFT_AMD_TILE2D");
+        return e ? atoi(e) : 1;
+    }();
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
         if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
+        if constexpr (KH > 1) {                                              \
+            if (tile2d && at && !big && !force_big) {                        \
+                if (tile2d == 2) FC_LAUNCH_TH2(KH, KW, 2, 64)                \
+                FC_LAUNCH_TH2(KH, KW, 1, 32)                                 \
+            }                                                                \
+        }                                                                    \
         if (tile11 && !(KH == 5 && KW == 1)) {                               \
             /* 5x1 excluded: vertical taps share no staged rows, so the  */ \
             /* extra workgroups just duplicate A slabs (39.1 vs 37.7 us) */ \
