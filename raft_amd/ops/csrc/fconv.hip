@@ -432,9 +432,19 @@ extern "C" void launch_fconv_nhwc_bf16(
     int mode, const void* h_state, const void* z_buf_in, void* z_buf_out,
     void* rh_out, int alltaps, int mtiles, int stride, hipStream_t s) {
     dim3 blk(256);
+    // big-tile threshold (block count of the 64x128 tiling above which the
+    // compute-efficient big tile beats small-tile grid saturation);
+    // RAFT_AMD_BIG_MIN overrides for probes.
+    static const long long big_min = [] {
+        const char* e = getenv("RAFT_AMD_BIG_MIN");
+        return e ? (long long)atoll(e) : (long long)512;
+    }();
     const long long big_blocks0 =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
-    const bool big0 = big_blocks0 >= 512;
+    // N < 128 leaves half the big tile's 128 output-channel columns idle
+    // (measured: encoder N=64 3x3s run 80.7 us big vs ~55 us small even at
+    // 3520 blocks) — the big tile requires a full N extent.
+    const bool big0 = big_blocks0 >= big_min && N >= 128;
     // alltaps (measured, tools/bench_fconv.py): wins for kh>1 on the small
     // tile (5x1: 43.6 -> 32.8 us; 3x3 convc2: 35.5 -> 28.4), loses on the
     // big tile (heads 3x3 N=512: 30.2 -> 39.8; LDS kills occupancy)
@@ -443,7 +453,7 @@ extern "C" void launch_fconv_nhwc_bf16(
     // MI355X has 256 CUs / 8 XCDs — batch-1 grids need the small tile.
     const long long big_blocks =
         (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
-    const bool big = big_blocks >= 512;
+    const bool big = big_blocks >= big_min && N >= 128;
     if (N <= 4 && mode == 0 && in2 == nullptr && n_off == 0 &&
         out_cstride == N) {
         const long long ncells = (long long)B * H * W;
@@ -484,7 +494,7 @@ extern "C" void launch_fconv_nhwc_bf16(
     // stride-2 (encoder) shapes: H/W here are OUTPUT dims; input = 2H x 2W
     if (stride == 2) {
         const long long big2 = (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
-        const bool bigt = big2 >= 512;
+        const bool bigt = big2 >= big_min && N >= 128;
         if (kh == 7 && kw == 7) {
             if (bigt) FC_LAUNCH2(7, 7, 2, 4, 64, 128)
             FC_LAUNCH2(7, 7, 1, 2, 32, 64)
