@@ -285,15 +285,18 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
                     const int d = tx - PBW;
                     const int par = (S == 2) ? (d & 1) : 0;
                     const int roff = (S == 2) ? ((d >> 1) + RLO) : (d + PBW);
-                    const int rslot = rsl + (TH == 2 ? (wm >> 4) : 0);
-                    const char* sA = sAbase +
-                        ((rslot * MT + mt) * PAR + par) * ABYTES;
                     short8 af[MI], bf[NJ];
 #pragma unroll
-                    for (int i = 0; i < MI; ++i)
+                    for (int i = 0; i < MI; ++i) {
+                        // per-lane slab row: the fragment's 16 m-positions
+                        // may span TH output rows (TH=4: 2 rows per wave)
+                        const int m_af = wm + i * 16 + (lane & 15);
+                        const int rslot = rsl + m_af / BMX;
+                        const char* sA = sAbase +
+                            ((rslot * MT + mt) * PAR + par) * ABYTES;
                         af[i] = *(const short8*)(
-                            sA + fswz(((TH == 2) ? 0 : wm + i * 16)
-                                      + (lane & 15) + roff, cb));
+                            sA + fswz(m_af % BMX + roff, cb));
+                    }
 #pragma unroll
                     for (int j = 0; j < NJ; ++j)
                         bf[j] = *(const short8*)(
@@ -324,9 +327,8 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             for (int r = 0; r < 4; ++r) {
                 const int m = wm + i * 16 + (lane >> 4) * 4 + r;
                 const int n = n0 + wn + j * 16 + (lane & 15);
-                const int yy2 = (TH == 2) ? y + (m >> 4) : y;
-                const int x = (TH == 2) ? (x0 + (m & 15))
-                                        : (x0 + mt * BM + m);
+                const int yy2 = y + m / BMX;
+                const int x = x0 + mt * BMX + m % BMX;
                 if (x >= W || n >= N || yy2 >= H) continue;
                 const long long p = ((long long)b * H + yy2) * W + x;
                 float v = acc[mt][i][j][r];
@@ -517,29 +519,33 @@ extern "C" void launch_fconv_nhwc_bf16(
         const char* e = getenv("RAFT_AMD_TILE11");
         return e ? atoi(e) : 1;
     }();
-#define FC_LAUNCH_TH2(KH, KW, NJ, BNv)                                      \
+#define FC_LAUNCH_THX(KH, KW, NJ, BNv, THv)                                  \
     {                                                                        \
-        dim3 grid(cdiv(N, BNv), ((H + 1) / 2) * cdiv(W, 16), B);             \
+        dim3 grid(cdiv(N, BNv),                                              \
+                  ((H + THv - 1) / THv) * cdiv(W, 32 / THv), B);             \
         hipLaunchKernelGGL(                                                  \
-            (fconv_nhwc_bf16_k<KH, KW, 1, NJ, true, 1, 1, 2>), grid, blk,    \
+            (fconv_nhwc_bf16_k<KH, KW, 1, NJ, true, 1, 1, THv>), grid, blk,  \
             0, s, FCONV_ARGS);                                               \
         return;                                                              \
     }
-    // 2-row tiles for the vertical-halo shapes (KH>1): measured 11.98 ->
-    // 11.77 ms/step on the headline config (3x3 24.5 -> 21.9 us, 5x1
-    // 37.7 -> 35.8). RAFT_AMD_TILE2D=0 disables, =2 selects the BN=64
-    // variant (measured worse: LDS-limited occupancy).
+    // 2D output tiles for the vertical-halo shapes (KH>1): vertical taps
+    // share staged row slabs. Measured on the headline config: TH=2
+    // 11.98 -> 11.77, TH=4 (4 rows x 8 cols) -> 10.64 ms/step; TH=4 also
+    // wins batch-8 (57.1 -> 56.6) and 1080p (37.0 -> 36.5).
+    // RAFT_AMD_TILE2D: 0 = off, 1 = TH2, 2 = TH2/BN64 (worse: LDS-limited
+    // occupancy), 4 = TH4 (default).
     static const int tile2d = [] {
         const char* e = getenv("RAFT_AMD_TILE2D");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 4;
     }();
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
         if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
         if constexpr (KH > 1) {                                              \
             if (tile2d && at && !big && !force_big) {                        \
-                if (tile2d == 2) FC_LAUNCH_TH2(KH, KW, 2, 64)                \
-                FC_LAUNCH_TH2(KH, KW, 1, 32)                                 \
+                if (tile2d == 2) FC_LAUNCH_THX(KH, KW, 2, 64, 2)             \
+                if (tile2d == 4) FC_LAUNCH_THX(KH, KW, 1, 32, 4)             \
+                FC_LAUNCH_THX(KH, KW, 1, 32, 2)                              \
             }                                                                \
         }                                                                    \
         if (tile11 && !(KH == 5 && KW == 1)) {                               \
