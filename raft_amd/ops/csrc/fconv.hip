@@ -544,6 +544,10 @@ extern "C" void launch_fconv_nhwc_bf16(
         if constexpr (KH > 1) {                                              \
             if (tile2d && at && !big && !force_big) {                        \
                 if (tile2d == 2) FC_LAUNCH_THX(KH, KW, 2, 64, 2)             \
+                if (tile2d == 8) {                                           \
+                    if (KH == 5 && KW == 1) FC_LAUNCH_THX(KH, KW, 1, 32, 8)  \
+                    FC_LAUNCH_THX(KH, KW, 1, 32, 4)                          \
+                }                                                            \
                 if (tile2d == 4) FC_LAUNCH_THX(KH, KW, 1, 32, 4)             \
                 FC_LAUNCH_THX(KH, KW, 1, 32, 2)                              \
             }                                                                \
