@@ -533,14 +533,22 @@ extern "C" void launch_fconv_nhwc_bf16(
     // 11.98 -> 11.77, TH=4 (4 rows x 8 cols) -> 10.64 ms/step; TH=4 also
     // wins batch-8 (57.1 -> 56.6) and 1080p (37.0 -> 36.5).
     // RAFT_AMD_TILE2D: 0 = off, 1 = TH2, 2 = TH2/BN64 (worse: LDS-limited
-    // occupancy), 4 = TH4 (default).
+    // occupancy), 4 = TH4 for KH>1, 5 = TH4 incl. 1x5 (default: 10.64 ->
+    // 10.29 ms/step — the 4-row tile wins even where it stages MORE halo,
+    // so the lever is occupancy, not just staging ratio), 8 = TH8 for 5x1
+    // (parity with TH4).
     static const int tile2d = [] {
         const char* e = getenv("RAFT_AMD_TILE2D");
-        return e ? atoi(e) : 4;
+        return e ? atoi(e) : 5;
     }();
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
         if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
+        if constexpr (KH == 1 && KW > 1) {                                   \
+            /* tile2d==5 probe: 4-row tiles for horizontal-tap shapes too */ \
+            if (tile2d == 5 && !big && !force_big)                           \
+                FC_LAUNCH_THX(KH, KW, 1, 32, 4)                              \
+        }                                                                    \
         if constexpr (KH > 1) {                                              \
             if (tile2d && at && !big && !force_big) {                        \
                 if (tile2d == 2) FC_LAUNCH_THX(KH, KW, 2, 64, 2)             \
@@ -548,7 +556,7 @@ extern "C" void launch_fconv_nhwc_bf16(
                     if (KH == 5 && KW == 1) FC_LAUNCH_THX(KH, KW, 1, 32, 8)  \
                     FC_LAUNCH_THX(KH, KW, 1, 32, 4)                          \
                 }                                                            \
-                if (tile2d == 4) FC_LAUNCH_THX(KH, KW, 1, 32, 4)             \
+                if (tile2d >= 4) FC_LAUNCH_THX(KH, KW, 1, 32, 4)             \
                 FC_LAUNCH_THX(KH, KW, 1, 32, 2)                              \
             }                                                                \
         }                                                                    \
