@@ -528,6 +528,15 @@ extern "C" void launch_fconv_nhwc_bf16(
             0, s, FCONV_ARGS);                                               \
         return;                                                              \
     }
+#define FC_LAUNCH_THX_MT2(KH, KW, THv)                                       \
+    {                                                                        \
+        dim3 grid(cdiv(N, 32),                                               \
+                  ((H + THv - 1) / THv) * cdiv(W, 2 * (32 / THv)), B);       \
+        hipLaunchKernelGGL(                                                  \
+            (fconv_nhwc_bf16_k<KH, KW, 1, 1, true, 2, 1, THv>), grid, blk,   \
+            0, s, FCONV_ARGS);                                               \
+        return;                                                              \
+    }
     // 2D output tiles for the vertical-halo shapes (KH>1): vertical taps
     // share staged row slabs. Measured on the headline config: TH=2
     // 11.98 -> 11.77, TH=4 (4 rows x 8 cols) -> 10.64 ms/step; TH=4 also
@@ -544,9 +553,16 @@ extern "C" void launch_fconv_nhwc_bf16(
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
         if (big || force_big) FC_LAUNCH(KH, KW, 2, 4, false, 1, 64, 128)     \
+        if constexpr (KH == 1 && KW == 1) {                                  \
+            /* tile2d==7 probe: 4-row tiles for 1x1 as well */               \
+            if (tile2d == 7 && !big && !force_big)                           \
+                FC_LAUNCH_THX(KH, KW, 1, 32, 4)                              \
+        }                                                                    \
         if constexpr (KH == 1 && KW > 1) {                                   \
             /* tile2d==5 probe: 4-row tiles for horizontal-tap shapes too */ \
-            if (tile2d == 5 && !big && !force_big)                           \
+            if (tile2d == 6 && !big && !force_big)                           \
+                FC_LAUNCH_THX_MT2(KH, KW, 4)                                 \
+            if ((tile2d == 5 || tile2d == 7) && !big && !force_big)          \
                 FC_LAUNCH_THX(KH, KW, 1, 32, 4)                              \
         }                                                                    \
         if constexpr (KH > 1) {                                              \
@@ -556,6 +572,7 @@ extern "C" void launch_fconv_nhwc_bf16(
                     if (KH == 5 && KW == 1) FC_LAUNCH_THX(KH, KW, 1, 32, 8)  \
                     FC_LAUNCH_THX(KH, KW, 1, 32, 4)                          \
                 }                                                            \
+                if (tile2d == 6) FC_LAUNCH_THX_MT2(KH, KW, 4)                \
                 if (tile2d >= 4) FC_LAUNCH_THX(KH, KW, 1, 32, 4)             \
                 FC_LAUNCH_THX(KH, KW, 1, 32, 2)                              \
             }                                                                \
