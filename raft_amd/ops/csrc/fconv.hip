@@ -545,10 +545,11 @@ extern "C" void launch_fconv_nhwc_bf16(
     // occupancy), 4 = TH4 for KH>1, 5 = TH4 incl. 1x5 (default: 10.64 ->
     // 10.29 ms/step — the 4-row tile wins even where it stages MORE halo,
     // so the lever is occupancy, not just staging ratio), 8 = TH8 for 5x1
-    // (parity with TH4).
+    // (parity with TH4), 6 = MT2-under-TH4 (loses: 10.75), 7 = TH4 for
+    // every non-big shape incl. 1x1 (default: 10.33 -> 10.21 ms/step).
     static const int tile2d = [] {
         const char* e = getenv("RAFT_AMD_TILE2D");
-        return e ? atoi(e) : 5;
+        return e ? atoi(e) : 7;
     }();
 #define FC_CASE(KH, KW)                                                      \
     if (kh == KH && kw == KW) {                                              \
