@@ -359,6 +359,84 @@ __global__ __launch_bounds__(256) void fconv_nhwc_bf16_k(
             }
 }
 
+
+// ------------------------------------------------------ stem conv (S2, C=8)
+// The encoder stems are 7x7 stride-2 over the 8-channel zero-padded image
+// (x8 buffer): the generic S2 path stages FC_BK=32-channel K-steps of
+// which only 8 carry data (measured 223 us/call after the small-tile
+// switch, 369 before). This kernel im2cols in LDS with K = taps*8: an
+// MFMA k-group's 8 contiguous k ARE one tap's 8 channels, so the whole
+// 7-row input tile (all taps) plus the full [52][32][8] weight tile fit
+// LDS -> ONE barrier, then 13 straight mfma_16x16x32 per wave.
+extern "C" __global__ __launch_bounds__(256) void fconv_stem_s2_k(
+    const __hip_bfloat16* __restrict__ in,   // [B, 2H, 2W, 8]
+    const __hip_bfloat16* __restrict__ wp,   // [49][N][8]
+    const float* __restrict__ bias,          // [N] or null
+    __hip_bfloat16* __restrict__ out,        // [B, H, W, N]
+    int H, int W, int N, int act) {
+    constexpr int KT = 49, KTP = 52;         // taps; padded to 13 k-chunks
+    constexpr int AROW = 72;                 // 69 staged input cols, padded
+    __shared__ __hip_bfloat16 sa[7 * AROW * 8];
+    __shared__ __hip_bfloat16 sb[KTP * 32 * 8];
+    const int b = blockIdx.z;
+    const int tiles = (W + 31) >> 5;
+    const int y = blockIdx.y / tiles;
+    const int x0 = (blockIdx.y % tiles) << 5;
+    const int n0 = blockIdx.x * 32;
+    const int tid = threadIdx.x;
+    const int Hi = 2 * H, Wi = 2 * W;
+    // A: rows 2y-2..2y+4, cols 2x0-2..2x0+66 (Conv2dTF even-input SAME:
+    // pad_beg = (7-2)/2 = 2), 8 ch = one 16-byte load per pixel
+    for (int e = tid; e < 7 * 69; e += 256) {
+        const int r = e / 69, c = e % 69;
+        const int yy = 2 * y + r - 2;
+        const int xx = 2 * x0 + c - 2;
+        uint4v v = {0, 0, 0, 0};
+        if (yy >= 0 && yy < Hi && xx >= 0 && xx < Wi)
+            v = *(const uint4v*)(
+                in + (((long long)b * Hi + yy) * Wi + xx) * 8);
+        *(uint4v*)(sa + ((size_t)r * AROW + c) * 8) = v;
+    }
+    // B: [52][32][8]; taps >= 49 and n >= N are zero (pad chunks vanish)
+    for (int e = tid; e < KTP * 32; e += 256) {
+        const int t = e / 32, n = e % 32;
+        uint4v v = {0, 0, 0, 0};
+        if (t < KT && n0 + n < N)
+            v = *(const uint4v*)(wp + ((size_t)t * N + n0 + n) * 8);
+        *(uint4v*)(sb + (size_t)e * 8) = v;
+    }
+    __syncthreads();
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = (wave >> 1) * 16;
+    const int wn = (wave & 1) * 16;
+    const int kg = lane >> 4;
+    const int m_af = wm + (lane & 15);
+    const int n_bf = wn + (lane & 15);
+    floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kc = 0; kc < 13; ++kc) {
+        const int t = kc * 4 + kg;
+        const int tc = t < KT ? t : KT - 1;   // pad taps: B is zero there,
+        const int dy = tc / 7, dx = tc % 7;   // clamp A address in-bounds
+        const short8 af = *(const short8*)(
+            sa + ((size_t)dy * AROW + 2 * m_af + dx) * 8);
+        const short8 bf = *(const short8*)(sb + ((size_t)t * 32 + n_bf) * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int mm = wm + (lane >> 4) * 4 + r;
+        const int nn = n0 + wn + (lane & 15);
+        const int x = x0 + mm;
+        if (x >= W || nn >= N) continue;
+        float v = acc[r];
+        if (bias) v += bias[nn];
+        out[(((long long)b * H + y) * W + x) * N + nn] =
+            (__hip_bfloat16)factivate(v, act);
+    }
+}
+
 // ----------------------------------------------------------- tiny-N conv
 // N <= 4 (the flow head's final 3x3 -> 2): an MFMA tile wastes 97% of its
 // columns. One wave per position: lanes stride over K = taps*Cin with N
@@ -495,6 +573,20 @@ extern "C" void launch_fconv_nhwc_bf16(
     }
     // stride-2 (encoder) shapes: H/W here are OUTPUT dims; input = 2H x 2W
     if (stride == 2) {
+        static const int stem = [] {
+            const char* e = getenv("RAFT_AMD_STEM");
+            return e ? atoi(e) : 1;
+        }();
+        if (stem && kh == 7 && kw == 7 && C1 == 8 && C2 == 0 &&
+            mode == EP_PLAIN && n_off == 0 && out_cstride == N &&
+            in1_off == 0 && in1_stride == 8 && N <= 128) {
+            dim3 grid(cdiv(N, 32), H * cdiv(W, 32), B);
+            hipLaunchKernelGGL(fconv_stem_s2_k, grid, blk, 0, s,
+                               (const __hip_bfloat16*)in1,
+                               (const __hip_bfloat16*)wp, bias,
+                               (__hip_bfloat16*)out, H, W, N, act);
+            return;
+        }
         const long long big2 = (long long)cdiv(N, 128) * H * cdiv(W, 64) * B;
         const bool bigt = big2 >= big_min && N >= 128;
         if (kh == 7 && kw == 7) {
