@@ -785,6 +785,112 @@ extern "C" __global__ __launch_bounds__(256) void fconv_smallk_lds_k(
     }
 }
 
+
+// small-C MFMA conv (stride 1): same im2col-in-LDS trick as the stem
+// kernel but for the motion encoder's convf1 (7x7 over the 2-channel
+// flow slice, K = taps*C = 98 -> 4 MFMA chunks). Three phases: stage the
+// raw kh x (32+kw-1) x C tile + the [n][K] weight image, barrier,
+// im2col-expand the 32 positions' windows into [pos][K] rows (LDS->LDS),
+// barrier, then kchunks straight mfma_16x16x32 per wave. Rows padded to
+// K+8 elements so the 16 m/n lanes of a fragment read stride 272 B
+// (4-bank rotation) instead of 256 B (single-bank pileup).
+#define SCM_KP 136                    // 128-padded K + 8-element row pad
+extern "C" __global__ __launch_bounds__(256) void fconv_smallc_mfma_k(
+    const __hip_bfloat16* __restrict__ in,    // [B,H,W,*] slice
+    int in_stride, int in_off,
+    const __hip_bfloat16* __restrict__ wp,    // [taps][N][C]
+    const float* __restrict__ bias,
+    __hip_bfloat16* __restrict__ out,         // [B,H,W,N]
+    int H, int W, int C, int N, int kh, int kw, int act, int kchunks) {
+    __shared__ __hip_bfloat16 sraw[9 * 40 * 4];
+    __shared__ __hip_bfloat16 sa[32 * SCM_KP];
+    __shared__ __hip_bfloat16 sb[32 * SCM_KP];
+    const int b = blockIdx.z;
+    const int tiles = (W + 31) >> 5;
+    const int y = blockIdx.y / tiles;
+    const int x0 = (blockIdx.y % tiles) << 5;
+    const int n0 = blockIdx.x * 32;
+    const int tid = threadIdx.x;
+    const int aw = 32 + kw - 1;
+    const int pb = kh / 2, pbw = kw / 2;
+    const int taps = kh * kw;
+    const int K = taps * C;
+    const int Kp = kchunks * 32;
+    for (int e = tid; e < kh * aw; e += 256) {
+        const int r = e / aw, cx = e % aw;
+        const int yy = y + r - pb, xx = x0 + cx - pbw;
+        const bool ok = (yy >= 0 && yy < H && xx >= 0 && xx < W);
+        const __hip_bfloat16* src =
+            in + (((long long)b * H + yy) * W + xx) * in_stride + in_off;
+        for (int c = 0; c < C; ++c)
+            sraw[e * C + c] = ok ? src[c] : (__hip_bfloat16)0.f;
+    }
+    // weights -> [n][K] rows; wp is contiguous in (n, c) per tap, so load
+    // 8 elements (= 8/C n-adjacent channel groups) per 16-byte chunk and
+    // scatter into the per-n rows
+    for (int e8 = tid; e8 < taps * 32 * C / 8; e8 += 256) {
+        const int per_tap = 32 * C / 8;
+        const int t = e8 / per_tap;
+        const int j = (e8 % per_tap) * 8;      // element offset within tap
+        __hip_bfloat16 tmp[8];
+        if (n0 * C + j + 8 <= N * C) {
+            *(uint4v*)tmp = *(const uint4v*)(
+                wp + ((size_t)t * N + n0) * C + j);
+        } else {
+            for (int u = 0; u < 8; ++u) {
+                const int n = n0 + (j + u) / C;
+                tmp[u] = n < N ? wp[((size_t)t * N + n) * C + (j + u) % C]
+                               : (__hip_bfloat16)0.f;
+            }
+        }
+        for (int u = 0; u < 8; ++u) {
+            const int n = (j + u) / C;
+            const int c = (j + u) % C;
+            sb[n * SCM_KP + t * C + c] = tmp[u];
+        }
+    }
+    // zero the K padding of the weight rows
+    for (int e = tid; e < 32 * (Kp - K); e += 256)
+        sb[(e / (Kp - K)) * SCM_KP + K + e % (Kp - K)] = (__hip_bfloat16)0.f;
+    __syncthreads();
+    for (int e = tid; e < 32 * Kp; e += 256) {
+        const int pos = e / Kp, k = e % Kp;
+        __hip_bfloat16 v = (__hip_bfloat16)0.f;
+        if (k < K) {
+            const int t = k / C, c = k % C;
+            v = sraw[((t / kw) * aw + pos + t % kw) * C + c];
+        }
+        sa[pos * SCM_KP + k] = v;
+    }
+    __syncthreads();
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wm = (wave >> 1) * 16;
+    const int wn = (wave & 1) * 16;
+    const int kg = lane >> 4;
+    const int m_af = wm + (lane & 15);
+    const int n_bf = wn + (lane & 15);
+    floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int kc = 0; kc < kchunks; ++kc) {
+        const short8 af = *(const short8*)(
+            sa + (size_t)m_af * SCM_KP + kc * 32 + kg * 8);
+        const short8 bf = *(const short8*)(
+            sb + (size_t)n_bf * SCM_KP + kc * 32 + kg * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int mm = wm + (lane >> 4) * 4 + r;
+        const int nn = n0 + wn + (lane & 15);
+        const int x = x0 + mm;
+        if (x >= W || nn >= N) continue;
+        float v = acc[r];
+        if (bias) v += bias[nn];
+        out[(((long long)b * H + y) * W + x) * N + nn] =
+            (__hip_bfloat16)factivate(v, act);
+    }
+}
+
 // 2D-tile small-C direct conv (stride 1): the per-position window staging
 // above re-reads every overlapped tap (convf1 7x7/C=2: 49 scattered 2-byte
 // loads per output pixel — 43.7 us/call measured). This version stages a
@@ -923,6 +1029,22 @@ extern "C" void launch_fconv_smallk_nhwc_bf16(
     // H, W are OUTPUT dims (input = conv_stride*H x conv_stride*W)
     const long long ncells = (long long)B * H * W;
     const int taps = kh * kw;
+    static const int scm = [] {
+        const char* e = getenv("RAFT_AMD_SMALLC_MFMA");
+        return e ? atoi(e) : 1;
+    }();
+    if (scm && conv_stride == 1 && C <= 4 && taps * C <= 128 && kh <= 9 &&
+        kw <= 9 && (N * C) % 8 == 0) {
+        const int kchunks = (taps * C + 31) / 32;
+        dim3 grid((unsigned)((N + 31) / 32),
+                  (unsigned)(H * ((W + 31) / 32)), (unsigned)B);
+        hipLaunchKernelGGL(fconv_smallc_mfma_k, grid, dim3(256), 0, s,
+                           (const __hip_bfloat16*)in, in_stride, in_off,
+                           (const __hip_bfloat16*)wp, bias,
+                           (__hip_bfloat16*)out, H, W, C, N, kh, kw, act,
+                           kchunks);
+        return;
+    }
     const int smem_tile =
         (kh * (16 + kw - 1) * C + taps * 64 * C) * (int)sizeof(float);
     if (conv_stride == 1 && C <= 4 && taps <= 81 && smem_tile <= 49152) {
