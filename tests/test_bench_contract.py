@@ -49,3 +49,25 @@ def test_bench_torchrun_world2_gloo():
     j = json.loads(jsons[0])
     assert j["n_gpus"] == 2
     assert j["config"]["parallelism"] == "dp2"
+
+
+def test_val_mode_torchrun_world2_gloo():
+    """DP-sharded eval: two gloo ranks split the seeds
+    and all-reduce the mean EPE; rank 0 prints one JSON result whose mean
+    aggregates BOTH ranks' shards (4 local batches each of 8 seeds)."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29741", "infer_raft.py", "-m", "val",
+         "--small", "--iters", "1", "--no-graph"],
+        cwd=REPO, capture_output=True, text=True, timeout=900, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    jsons = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(jsons) == 1, jsons
+    j = json.loads(jsons[0])
+    assert j["world"] == 2
+    assert len(j["epe_per_batch"]) == 4      # rank 0's shard of 8 seeds
+    assert 0 < j["epe_mean"] < 50
