@@ -193,7 +193,6 @@ def test_jpeg_noise_quality_monotone_and_shape():
 
 def test_augment_pair_shared_params_and_flow_consistency():
     from raft_amd.data.dataflow import augment_pair
-    from raft_amd.engine.trainer import epe
     # the same seed must give identical augmentation twice (shared RNG)
     im1 = torch.rand(1, 3, 64, 96)
     im2 = torch.rand(1, 3, 64, 96)
