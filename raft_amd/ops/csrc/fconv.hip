@@ -20,14 +20,22 @@
 //     (n_off / out_cstride), which is how the per-iteration GRU input
 //     buffer x = [ctx | motion | flow] is assembled without copies.
 //
-// Geometry: 256 threads = 4 waves (2m x 2n); tile 64x128 (big) or 32x64
-// (small — batch-1 grid saturation), selected by grid size. K-loop: per
-// (kernel row, BK=32 channel step) stage the halo'd input slab(s) and
-// weight tiles (AT variant: all taps per step), one barrier pair, then
+// Geometry: 256 threads = 4 waves (2m x 2n). Tile selection (all
+// A/B-measured, see profiles/r01_final_optimization_pass.md): 64x128
+// (big) only when the grid is large AND N >= 128 fills the tile's
+// columns; otherwise 2D tiles of 4 output rows x 8 cols (TH=4 — at
+// batch-1 grids per-CU concurrency beats per-wave MFMA efficiency, and
+// vertical taps share the staged row slabs). K-loop: per (kernel row,
+// BK=32 channel step) stage the halo'd input slab(s) and weight tiles
+// (AT variant: all taps per step), one barrier pair, then
 // shifted-LDS-read MFMA groups; stride-2 stages even/odd input columns as
 // parity slabs. Interior tiles take unguarded staging fast paths (per-
-// element guarded loads serialize — guide §5 trap 4c). Every structural
-// choice here is A/B-measured: see tools/bench_fconv.py and profiles/.
+// element guarded loads serialize — guide §5 trap 4c). Small-Cin shapes
+// bypass this kernel entirely: 7x7-S2 stems and the 7x7/C=2 convf1 run
+// as im2col-in-LDS MFMA GEMMs with K = taps*C (one barrier per block).
+// Every structural choice here is A/B-measured: see tools/bench_fconv.py
+// and profiles/; losers stay selectable via the RAFT_AMD_* switches
+// documented in ARCHITECTURE.md.
 
 #include "common.h"
 #include <hip/hip_bf16.h>
