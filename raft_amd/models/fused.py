@@ -103,16 +103,37 @@ class FusedBasicUpdate:
         self.fh2 = _FC(ub.flow_head.conv2)
         self.m2 = _FC(ub.mask[2], scale=0.25)        # fold the 0.25 scale
         self.ctx_dim = ctx_dim
+        self.side_stream = None                      # set by FusedRaft.run
+        self._ev_fork = torch.cuda.Event()
+        self._ev_join = torch.cuda.Event()
 
     def __call__(self, hip, net, x_buf, corr_pad, coords1, final=True):
         # motion encoder (model_utils.py:110-119). The flow channels of
         # x_buf were already written by the lookup kernel (fused flow out).
+        # The corr branch (c1->c2) and flow branch (f1->f2) are
+        # independent until cv and each is latency-bound, so they run on
+        # two streams (fork after the lookup, join before cv).
         ctx = self.ctx_dim
-        cor = self.c2(hip, self.c1(hip, corr_pad))
-        flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
-                                self.f1.kh, self.f1.kw, ACT_RELU,
-                                ctx + 126, 2, 1)
-        flo = self.f2(hip, flo1)
+        side = self.side_stream
+        if side is not None and not torch.cuda.is_current_stream_capturing():
+            cur = torch.cuda.current_stream()
+            self._ev_fork.record(cur)
+            side.wait_event(self._ev_fork)
+            with torch.cuda.stream(side):
+                flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                        self.f1.kh, self.f1.kw, ACT_RELU,
+                                        ctx + 126, 2, 1)
+                flo = self.f2(hip, flo1)
+                self._ev_join.record(side)
+            cor = self.c2(hip, self.c1(hip, corr_pad))
+            cur.wait_event(self._ev_join)
+            flo.record_stream(cur)
+        else:
+            cor = self.c2(hip, self.c1(hip, corr_pad))
+            flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                    self.f1.kh, self.f1.kw, ACT_RELU,
+                                    ctx + 126, 2, 1)
+            flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 126 ch
         # SepConvGRU (model_utils.py:138-156)
         net = self.gru1(hip, net, x_buf)
@@ -150,15 +171,33 @@ class FusedSmallUpdate:
         self.fh1 = _FC(ub.flow_head.conv1)
         self.fh2 = _FC(ub.flow_head.conv2)
         self.ctx_dim = ctx_dim
+        self.side_stream = None                      # set by FusedRaft.run
+        self._ev_fork = torch.cuda.Event()
+        self._ev_join = torch.cuda.Event()
         # x = [inp(ctx) | motion(80) | flow(2)]; motion encoder out = 80
 
     def __call__(self, hip, net, x_buf, corr_pad, coords1, final=True):
         ctx = self.ctx_dim
-        cor = self.c1(hip, corr_pad)
-        flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
-                                self.f1.kh, self.f1.kw, ACT_RELU,
-                                ctx + 80, 2, 1)
-        flo = self.f2(hip, flo1)
+        side = self.side_stream
+        if side is not None and not torch.cuda.is_current_stream_capturing():
+            cur = torch.cuda.current_stream()
+            self._ev_fork.record(cur)
+            side.wait_event(self._ev_fork)
+            with torch.cuda.stream(side):
+                flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                        self.f1.kh, self.f1.kw, ACT_RELU,
+                                        ctx + 80, 2, 1)
+                flo = self.f2(hip, flo1)
+                self._ev_join.record(side)
+            cor = self.c1(hip, corr_pad)
+            cur.wait_event(self._ev_join)
+            flo.record_stream(cur)
+        else:
+            cor = self.c1(hip, corr_pad)
+            flo1 = hip.fconv_smallk(x_buf, self.f1.wp, self.f1.bias,
+                                    self.f1.kh, self.f1.kw, ACT_RELU,
+                                    ctx + 80, 2, 1)
+            flo = self.f2(hip, flo1)
         self.cv(hip, cor, flo, ACT_RELU, out=x_buf, n_off=ctx)   # 80 ch
         net = self.gru(hip, net, x_buf)
         h1 = self.fh1(hip, net)
@@ -370,6 +409,7 @@ class FusedRaft:
         if side is None:
             side = torch.cuda.Stream()
             self._side_stream = side
+        self.update.side_stream = side   # motion-encoder branch overlap
 
         if self.fuse_enc:
             # in-repo encoders: images -> physical NHWC padded to 8 ch
