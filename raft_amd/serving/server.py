@@ -99,6 +99,40 @@ def create_app(model=None, iters: Optional[int] = None,
             request_errors.inc()
             raise
 
+    @app.post("/flow_batch")
+    async def flow_batch(request: Request, iters: Optional[int] = None):
+        """Batched flow with per-pair dynamic shapes. Body framing:
+        [uint32-le n_pairs] then per pair
+        [uint32 len(png1)][png1][uint32 len(png2)][png2].
+        Same-shape pairs are grouped into one engine call
+        (run_mixed_batch — BASELINE config 5 semantics); the response is
+        the concatenation of one .flo record per pair, input order."""
+        from raft_amd.engine.inference import run_mixed_batch
+        t0 = time.perf_counter()
+        try:
+            body = await request.body()
+            (n_pairs,) = struct.unpack_from("<I", body, 0)
+            off = 4
+            pairs = []
+            for _ in range(n_pairs):
+                (n1,) = struct.unpack_from("<I", body, off)
+                im1 = _decode_png_bytes(body[off + 4:off + 4 + n1])
+                off += 4 + n1
+                (n2,) = struct.unpack_from("<I", body, off)
+                im2 = _decode_png_bytes(body[off + 4:off + 4 + n2])
+                off += 4 + n2
+                pairs.append((im1, im2))
+            flows = run_mixed_batch(engine, pairs, iters=iters)
+            payload = b"".join(
+                _flo_bytes(f[0].float().permute(1, 2, 0).cpu().numpy())
+                for f in flows)
+            requests_total.inc()
+            latency.observe(time.perf_counter() - t0)
+            return Response(payload, media_type="application/octet-stream")
+        except Exception:
+            request_errors.inc()
+            raise
+
     return app
 
 

@@ -62,3 +62,34 @@ def test_metrics(client):
     r = client.get("/metrics")
     assert r.status_code == 200
     assert b"raft_requests_total" in r.content
+
+
+def test_flow_batch_endpoint_mixed_shapes():
+    import struct
+    import numpy as np
+    import torch
+    from starlette.testclient import TestClient
+    from raft_amd import RAFT, RaftConfig
+    from raft_amd.serving.server import create_app
+    from raft_amd.data.imageio import encode_png
+
+    app = create_app(RAFT(RaftConfig(small=True)), iters=2)
+    client = TestClient(app)
+    rng = np.random.default_rng(5)
+    shapes = [(48, 64), (32, 48), (48, 64)]       # two groups, order check
+    body = struct.pack("<I", len(shapes))
+    for h, w in shapes:
+        p1 = encode_png(rng.integers(0, 256, (h, w, 3), dtype=np.uint8))
+        p2 = encode_png(rng.integers(0, 256, (h, w, 3), dtype=np.uint8))
+        body += struct.pack("<I", len(p1)) + p1
+        body += struct.pack("<I", len(p2)) + p2
+    r = client.post("/flow_batch", content=body)
+    assert r.status_code == 200
+    data = r.content
+    off = 0
+    for h, w in shapes:                           # .flo records, input order
+        assert data[off:off + 4] == b"PIEH"
+        fw, fh = struct.unpack_from("<ii", data, off + 4)
+        assert (fh, fw) == (h, w)
+        off += 12 + fh * fw * 2 * 4
+    assert off == len(data)
