@@ -21,7 +21,7 @@ import torch
 
 try:
     from fastapi import FastAPI, Request, Response
-    from prometheus_client import (Counter, Histogram,
+    from prometheus_client import (Counter, Histogram, CollectorRegistry,
                                    generate_latest, CONTENT_TYPE_LATEST)
     _HAVE_SERVING = True
 except ImportError:  # pragma: no cover
@@ -57,9 +57,15 @@ def create_app(model=None, iters: Optional[int] = None,
     engine = InferenceEngine(model, iters=iters, dtype=dtype)
 
     app = FastAPI(title="raft_amd flow service")
-    requests_total = Counter("raft_requests_total", "flow requests served")
-    request_errors = Counter("raft_request_errors_total", "failed requests")
-    latency = Histogram("raft_request_seconds", "end-to-end request latency")
+    # per-app registry: repeated create_app() in one process (tests,
+    # multi-model serving) must not collide in the global registry
+    registry = CollectorRegistry()
+    requests_total = Counter("raft_requests_total", "flow requests served",
+                             registry=registry)
+    request_errors = Counter("raft_request_errors_total", "failed requests",
+                             registry=registry)
+    latency = Histogram("raft_request_seconds", "end-to-end request latency",
+                        registry=registry)
 
     @app.get("/healthz")
     def healthz():
@@ -68,7 +74,8 @@ def create_app(model=None, iters: Optional[int] = None,
 
     @app.get("/metrics")
     def metrics():
-        return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+        return Response(generate_latest(registry),
+                        media_type=CONTENT_TYPE_LATEST)
 
     @app.post("/flow")
     async def flow(request: Request, fmt: str = "flo",
