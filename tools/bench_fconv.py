@@ -89,6 +89,28 @@ def main():
 
         print(f"gru {tag}: {bench(gru_pass):.1f} us/pass")
 
+    # small-Cin specialized kernels (dispatch selected by env at import —
+    # rerun the script under RAFT_AMD_SMALLC_MFMA=0 / RAFT_AMD_STEM=0 to
+    # time the fallback paths for the same shapes)
+    print("-- small-Cin kernels (env-dispatched; see RAFT_AMD_* switches)")
+    for name, cin, n, kh, kw, slice_dim in (
+            ("convf1 7x7/C2", 2, 128, 7, 7, 256),
+            ("small f1 7x7/C2", 2, 64, 7, 7, 128)):
+        xs = torch.randn(B, H, W, slice_dim, device=dev).to(torch.bfloat16)
+        wp = torch.randn(kh * kw, n, cin, device=dev).to(torch.bfloat16) * .1
+        bias = torch.zeros(n, device=dev)
+        t = bench(lambda: hip.fconv_smallk(xs, wp, bias, kh, kw, 1,
+                                           slice_dim - cin, cin, 1))
+        print(f"{name:<18} {t:>7.1f} us")
+    # stem: 7x7 stride-2 over the 8-channel padded image (fnet batch 2)
+    Hs, Ws = 220, 512
+    x8 = torch.randn(2, 2 * Hs, 2 * Ws, 8, device=dev).to(torch.bfloat16)
+    wp8 = torch.randn(49, 64, 8, device=dev).to(torch.bfloat16) * 0.1
+    b8 = torch.zeros(64, device=dev)
+    t = bench(lambda: hip.fconv_plain(x8, None, wp8, b8, 7, 7, 1, None,
+                                      0, 0, 0, -1, -1, 2, None), iters=20)
+    print(f"{'stem 7x7 S2/C8':<18} {t:>7.1f} us")
+
 
 if __name__ == "__main__":
     main()
