@@ -279,3 +279,15 @@ def test_png_decoder_rejects_garbage():
     from raft_amd.data.imageio import decode_png
     with pytest.raises(ValueError):
         decode_png(b"not a png at all")
+
+
+def test_prefetcher_cpu_passthrough_order():
+    from raft_amd.engine.inference import Prefetcher
+    batches = [(torch.full((1, 3, 8, 8), float(i)),
+                torch.full((1, 3, 8, 8), float(-i))) for i in range(5)]
+    out = list(Prefetcher(batches, torch.device("cpu"),
+                          dtype=torch.float32))
+    assert len(out) == 5
+    for i, (a, b) in enumerate(out):
+        assert float(a.flatten()[0]) == i
+        assert float(b.flatten()[0]) == -i
